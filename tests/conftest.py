@@ -1,15 +1,33 @@
+import asyncio
+import inspect
 import os
 import sys
 
 import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an MI355X GPU (run with -m gpu on a GPU box)"
     )
+    config.addinivalue_line(
+        "markers", "asyncio: run the test function in a fresh asyncio event loop"
+    )
+
+
+@pytest.hookimpl(tryfirst=True)
+def pytest_pyfunc_call(pyfuncitem):
+    """Minimal asyncio runner (pytest-asyncio is not in this image)."""
+    func = pyfuncitem.obj
+    if inspect.iscoroutinefunction(func):
+        kwargs = {name: pyfuncitem.funcargs[name]
+                  for name in pyfuncitem._fixtureinfo.argnames}
+        asyncio.run(func(**kwargs))
+        return True
+    return None
 
 
 def pytest_collection_modifyitems(config, items):

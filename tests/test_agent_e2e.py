@@ -1,0 +1,242 @@
+"""End-to-end orchestrator tests with scripted engines through the production
+path (the reference's tier-2 fake-backend strategy)."""
+
+import asyncio
+import json
+
+import pytest
+
+from quoracle_amd.engine.fake import FakeEngine
+
+from helpers import (IDLE, POOL2, action_json, make_manager, make_runtime,
+                     wait_until)
+
+
+@pytest.mark.asyncio
+async def test_task_creates_root_agent_and_decides():
+    engine = FakeEngine(default_response=IDLE)
+    todo = action_json("todo", {"items": [{"content": "step 1", "state": "todo"}]})
+    for m in POOL2:
+        engine.push_response(m, todo)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("do something", "default")
+    root_id = result["root_agent_id"]
+    assert runtime.registry.alive(root_id)
+
+    ok = await wait_until(
+        lambda: runtime.registry.lookup(root_id) is not None
+        and runtime.registry.lookup(root_id).actor.state.todos)
+    assert ok, "todo action never executed"
+    actor = runtime.registry.lookup(root_id).actor
+    assert actor.state.todos[0]["content"] == "step 1"
+    # decision + result entries landed in every model's history
+    for m in POOL2:
+        types = [e["type"] for e in actor.state.model_histories[m]]
+        assert "decision" in types
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_refinement_round_on_disagreement():
+    engine = FakeEngine(default_response=IDLE)
+    # round 1: disagreement on exact-match param -> two clusters
+    engine.push_response("fake-a", action_json("file_read", {"path": "/a"}))
+    engine.push_response("fake-b", action_json("file_read", {"path": "/b"}))
+    # round 2: both converge
+    engine.push_response("fake-a", action_json("file_read", {"path": "/tmp/x"}))
+    engine.push_response("fake-b", action_json("file_read", {"path": "/tmp/x"}))
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("read the file", "default")
+    root_id = result["root_agent_id"]
+
+    def got_result():
+        entry = runtime.registry.lookup(root_id)
+        if entry is None:
+            return False
+        h = entry.actor.state.model_histories["fake-a"]
+        return any(e["type"] == "result" for e in h)
+
+    assert await wait_until(got_result)
+    # refinement prompt was sent: at least 4 generate calls before defaults
+    refinement_calls = [c for c in engine.calls
+                        if any("Consensus Refinement" in m["content"]
+                               for m in c.messages)]
+    assert refinement_calls, "no refinement round happened"
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_round1_unanimity_required():
+    """Two models agreeing on action but differing on an exact-match param
+    must NOT settle in round 1."""
+    engine = FakeEngine(default_response=IDLE)
+    engine.push_response("fake-a", action_json("send_message",
+                                               {"to": "parent", "content": "hello"}))
+    engine.push_response("fake-b", action_json("send_message",
+                                               {"to": "children", "content": "hello"}))
+    engine.push_response("fake-a", action_json("send_message",
+                                               {"to": "parent", "content": "hello"}))
+    engine.push_response("fake-b", action_json("send_message",
+                                               {"to": "parent", "content": "hello"}))
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("report", "default")
+    root_id = result["root_agent_id"]
+
+    # root send_message to parent routes to the user mailbox topic
+    q = runtime.bus.subscribe(f"tasks:{result['task_id']}:messages")
+    event = await asyncio.wait_for(q.get(), timeout=5)
+    assert event.payload["to"] == "user"
+    assert event.payload["content"] == "hello"
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_spawn_child_and_message_flow():
+    engine = FakeEngine(default_response=IDLE)
+    spawn = action_json("spawn_child", {
+        "task_description": "analyze the data",
+        "success_criteria": "a summary exists",
+        "immediate_context": "data is in /tmp",
+        "approach_guidance": "be quick",
+        "profile": "default",
+        "budget": "10.00",
+    }, wait=True)
+    for m in POOL2:
+        engine.push_response(m, spawn)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("parent task", "default",
+                                       budget_limit=100.0)
+    root_id = result["root_agent_id"]
+
+    def child_running():
+        entry = runtime.registry.lookup(root_id)
+        if not entry:
+            return False
+        kids = runtime.registry.children_of(root_id)
+        return bool(kids) and all(
+            runtime.registry.lookup(k) is not None for k in kids)
+
+    assert await wait_until(child_running)
+    root = runtime.registry.lookup(root_id).actor
+    kids = runtime.registry.children_of(root_id)
+    assert len(kids) == 1
+    child = runtime.registry.lookup(kids[0]).actor
+    # escrow locked on parent
+    assert root.state.budget_committed == pytest.approx(10.0)
+    assert child.state.budget_allocated == pytest.approx(10.0)
+    assert child.state.profile == "default"
+    # child got the initial task message in history eventually
+    assert await wait_until(lambda: any(
+        "analyze the data" in str(e.get("content", ""))
+        for e in child.state.model_histories["fake-a"]))
+    # child_spawned notification reached the parent history
+    assert await wait_until(lambda: any(
+        "is now running" in str(e.get("content", ""))
+        for e in root.state.model_histories["fake-a"]))
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_dismiss_child_releases_escrow_and_terminates_tree():
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("t", "default", budget_limit=100.0)
+    root_id = result["root_agent_id"]
+    root = runtime.registry.lookup(root_id).actor
+
+    spawn_result = await runtime.supervisor.spawn_child_action(root, {
+        "task_description": "x", "success_criteria": "y",
+        "immediate_context": "z", "approach_guidance": "w",
+        "profile": "default", "budget": "20.00"})
+    child_id = spawn_result["child_id"]
+    assert await wait_until(lambda: runtime.registry.alive(child_id))
+    assert root.state.budget_committed == pytest.approx(20.0)
+
+    await runtime.supervisor.dismiss_child_action(root, child_id, "done")
+    assert await wait_until(lambda: not runtime.registry.alive(child_id))
+    assert await wait_until(
+        lambda: root.state.budget_committed == pytest.approx(0.0))
+    assert child_id not in root.state.children
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_consensus_stall_notifies_user():
+    engine = FakeEngine(default_response="THIS IS NOT JSON")
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("t", "default")
+    q = runtime.bus.subscribe(f"tasks:{result['task_id']}:messages")
+    event = await asyncio.wait_for(q.get(), timeout=10)
+    assert "stalled" in event.payload["content"]
+    await manager.supervisor.terminate_tree(result["root_agent_id"])
+
+
+@pytest.mark.asyncio
+async def test_action_gate_blocks_uncapable_action():
+    from quoracle_amd.governance.profiles import Profile
+    engine = FakeEngine(default_response=IDLE)
+    shell = action_json("execute_shell", {"command": "echo hi"})
+    for m in POOL2:
+        engine.push_response(m, shell)
+    manager, runtime = make_manager(engine)
+    runtime.profiles.put(Profile(name="restricted", model_pool=POOL2,
+                                 capability_groups=[]))  # no local_execution
+    result = await manager.create_task("t", "restricted")
+    root_id = result["root_agent_id"]
+    root = runtime.registry.lookup(root_id).actor
+    assert await wait_until(lambda: any(
+        "action_not_allowed" in str(e.get("content", ""))
+        for e in root.state.model_histories["fake-a"]))
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_pause_and_restore_task():
+    engine = FakeEngine(default_response=IDLE)
+    todo = action_json("todo", {"items": [{"content": "persist me",
+                                           "state": "pending"}]})
+    for m in POOL2:
+        engine.push_response(m, todo)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("t", "default")
+    task_id, root_id = result["task_id"], result["root_agent_id"]
+    root = runtime.registry.lookup(root_id).actor
+    assert await wait_until(lambda: root.state.todos)
+
+    await manager.pause_task(task_id)
+    assert not runtime.registry.alive(root_id)
+    assert runtime.store.get_task(task_id)["status"] == "paused"
+
+    restore = await manager.restore_task(task_id)
+    assert root_id in restore["restored"]
+    restored = runtime.registry.lookup(root_id).actor
+    assert restored.state.todos[0]["content"] == "persist me"
+    # history survived the round trip
+    assert any(e["type"] == "decision"
+               for e in restored.state.model_histories["fake-a"])
+    await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_boot_revival():
+    engine = FakeEngine(default_response=IDLE)
+    from quoracle_amd.persistence.store import Store
+    store = Store(":memory:")
+    manager, runtime = make_manager(engine, store=store)
+    result = await manager.create_task("t", "default")
+    task_id, root_id = result["task_id"], result["root_agent_id"]
+    assert await wait_until(lambda: runtime.registry.lookup(root_id) is not None
+                            and runtime.registry.lookup(root_id).actor.steps_completed >= 0)
+    # hard "crash": drop the runtime without pausing
+    entry = runtime.registry.lookup(root_id)
+    entry.actor._running = False
+    entry.actor._task.cancel()
+    runtime.registry.unregister(root_id)
+
+    # new runtime over the same store
+    manager2, runtime2 = make_manager(FakeEngine(default_response=IDLE),
+                                      store=store)
+    results = await manager2.restore_running_tasks()
+    assert root_id in results[task_id]["restored"]
+    assert runtime2.registry.alive(root_id)
+    await manager2.supervisor.terminate_tree(root_id)
