@@ -1,0 +1,283 @@
+// Paged-KV attention kernels (decode + chunked prefill) for MI355X.
+//
+// v1 structure: correctness-first VALU kernels with online softmax, paged KV
+// gather, GQA sharing (decode reads each K/V page once per KV head and serves
+// all its query heads), fp32 accumulation, bf16 storage.  The MFMA-tiled
+// prefill (guide §5.5 T10/T14/T16 structure) replaces phase A when profiling
+// shows prefill attention on the critical path; decode is HBM-bound on KV
+// reads, which this layout already streams.
+//
+// KV cache layout: [num_blocks, Hkv, BLOCK_SIZE, D] bf16 — a (block, head)
+// panel is BLOCK_SIZE*D*2 bytes contiguous, so both phases read contiguous
+// rows.
+
+#include "common.h"
+
+#define MAX_GQ 8      // query heads per KV head (llama3-8b: 4)
+#define CHUNK 128     // KV tokens processed per software chunk
+#define DECODE_BLOCK 128
+
+// ---------------------------------------------------------------------------
+// paged_attn_decode: one query token per sequence.
+//   q:    [B, Hq, D] bf16      out: [B, Hq, D] bf16
+//   kc/vc:[NB, Hkv, BS, D] bf16
+//   bt:   [B, MAXB] int32 block tables;  ctx: [B] int32 context lengths
+// Grid: (B, Hkv); block: DECODE_BLOCK threads (D <= DECODE_BLOCK).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_decode_kernel(bf16 *__restrict__ out, const bf16 *__restrict__ q,
+                         const bf16 *__restrict__ kc,
+                         const bf16 *__restrict__ vc,
+                         const int *__restrict__ bt,
+                         const int *__restrict__ ctx,
+                         float scale, int Hq, int Hkv, int D, int BS,
+                         int MAXB, int GQ) {
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int len = ctx[b];
+  if (len <= 0) return;
+
+  __shared__ float q_s[MAX_GQ * 128];
+  __shared__ float p_s[MAX_GQ][CHUNK];
+  __shared__ float scratch[8];
+
+  // Load the GQ query heads that share this KV head into LDS (fp32).
+  for (int i = tid; i < GQ * D; i += blockDim.x) {
+    int g = i / D, d = i % D;
+    q_s[g * D + d] = bf2f(q[((long)b * Hq + hk * GQ + g) * D + d]) * scale;
+  }
+  __syncthreads();
+
+  float m[MAX_GQ], l[MAX_GQ], acc[MAX_GQ];
+#pragma unroll
+  for (int g = 0; g < MAX_GQ; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+    acc[g] = 0.f;
+  }
+
+  const long panel_stride = (long)Hkv * BS * D;  // one cache block
+  for (int start = 0; start < len; start += CHUNK) {
+    const int clen = min(CHUNK, len - start);
+
+    // Phase A: thread -> KV token; dot against all GQ query heads.
+    if (tid < clen) {
+      const int token = start + tid;
+      const long blk = bt[(long)b * MAXB + token / BS];
+      const bf16 *krow =
+          kc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+      float dots[MAX_GQ];
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) dots[g] = 0.f;
+      for (int d8 = 0; d8 < D / 8; ++d8) {
+        uint4 kv = reinterpret_cast<const uint4 *>(krow)[d8];
+        float kf[8];
+        unpack_bf16x2(kv.x, kf[0], kf[1]);
+        unpack_bf16x2(kv.y, kf[2], kf[3]);
+        unpack_bf16x2(kv.z, kf[4], kf[5]);
+        unpack_bf16x2(kv.w, kf[6], kf[7]);
+        // static bounds + guard keep dots[] in registers (guide §5.4 rule 20)
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          const float *qg = q_s + g * D + d8 * 8;
+#pragma unroll
+          for (int k = 0; k < 8; ++k) dots[g] = fmaf(kf[k], qg[k], dots[g]);
+        }
+      }
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) {
+        if (g >= GQ) break;
+        p_s[g][tid] = dots[g];
+      }
+    }
+    __syncthreads();
+
+    // Online-softmax update per query head (all threads compute identical
+    // reductions; block_sum/max return the value to every thread).
+    for (int g = 0; g < GQ; ++g) {
+      float mine = (tid < clen) ? p_s[g][tid] : -INFINITY;
+      float cmax = block_max(mine, scratch);
+      float mn = fmaxf(m[g], cmax);
+      float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+      float p = (tid < clen) ? __expf(p_s[g][tid] - mn) : 0.f;
+      if (tid < clen) p_s[g][tid] = p;
+      float psum = block_sum(p, scratch);
+      l[g] = l[g] * alpha + psum;
+      acc[g] *= alpha;
+      m[g] = mn;
+    }
+    __syncthreads();
+
+    // Phase B: thread -> output dim; accumulate P·V over the chunk.
+    if (tid < D) {
+      for (int i = 0; i < clen; ++i) {
+        const int token = start + i;
+        const long blk = bt[(long)b * MAXB + token / BS];
+        const bf16 *vrow =
+            vc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+        const float v = bf2f(vrow[tid]);
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          acc[g] = fmaf(p_s[g][i], v, acc[g]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (tid < D) {
+#pragma unroll
+    for (int g = 0; g < MAX_GQ; ++g) {
+      if (g >= GQ) break;
+      const float o = (l[g] > 0.f) ? acc[g] / l[g] : 0.f;
+      out[((long)b * Hq + hk * GQ + g) * D + tid] = f2bf(o);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// paged_attn_prefill: chunked causal attention for new (suffix) tokens over
+// the full paged context (cached prefix + the suffix itself, already
+// appended to the cache by kv_append).
+//
+// Work decomposition: one block per (tile, qhead); a tile is QT=16
+// consecutive query tokens of one sequence.
+//   q:        [Tq, Hq, D] bf16 (all new tokens of the batch, seq-major)
+//   out:      [Tq, Hq, D] bf16
+//   tile_q0:  [ntiles] int32 — row index of the tile's first token in q
+//   tile_qn:  [ntiles] int32 — number of q tokens in this tile (<= 16)
+//   tile_seq: [ntiles] int32 — sequence index (selects block table)
+//   tile_pos0:[ntiles] int32 — absolute position of the tile's first token
+// Grid: (ntiles, Hq); block: 128.
+// ---------------------------------------------------------------------------
+#define QT 16
+
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_prefill_kernel(bf16 *__restrict__ out, const bf16 *__restrict__ q,
+                          const bf16 *__restrict__ kc,
+                          const bf16 *__restrict__ vc,
+                          const int *__restrict__ bt,
+                          const int *__restrict__ tile_q0,
+                          const int *__restrict__ tile_qn,
+                          const int *__restrict__ tile_seq,
+                          const int *__restrict__ tile_pos0,
+                          float scale, int Hq, int Hkv, int D, int BS,
+                          int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+
+  __shared__ float q_s[QT * 128];
+  __shared__ float p_s[QT][CHUNK];
+  __shared__ float scratch[8];
+
+  for (int i = tid; i < qn * D; i += blockDim.x) {
+    int qi = i / D, d = i % D;
+    q_s[qi * D + d] = bf2f(q[((long)(q0 + qi) * Hq + h) * D + d]) * scale;
+  }
+  __syncthreads();
+
+  float m[QT], l[QT], acc[QT];
+#pragma unroll
+  for (int qi = 0; qi < QT; ++qi) {
+    m[qi] = -INFINITY;
+    l[qi] = 0.f;
+    acc[qi] = 0.f;
+  }
+
+  const int kv_limit = pos0 + qn;  // last q row attends to positions < pos0+qn
+  const long panel_stride = (long)Hkv * BS * D;
+
+  for (int start = 0; start < kv_limit; start += CHUNK) {
+    const int clen = min(CHUNK, kv_limit - start);
+
+    if (tid < clen) {
+      const int token = start + tid;
+      const long blk = bt[(long)seq * MAXB + token / BS];
+      const bf16 *krow =
+          kc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+      float dots[QT];
+#pragma unroll
+      for (int qi = 0; qi < QT; ++qi) dots[qi] = 0.f;
+      for (int d8 = 0; d8 < D / 8; ++d8) {
+        uint4 kv = reinterpret_cast<const uint4 *>(krow)[d8];
+        float kf[8];
+        unpack_bf16x2(kv.x, kf[0], kf[1]);
+        unpack_bf16x2(kv.y, kf[2], kf[3]);
+        unpack_bf16x2(kv.z, kf[4], kf[5]);
+        unpack_bf16x2(kv.w, kf[6], kf[7]);
+#pragma unroll
+        for (int qi = 0; qi < QT; ++qi) {
+          if (qi >= qn) break;
+          const float *qg = q_s + qi * D + d8 * 8;
+#pragma unroll
+          for (int k = 0; k < 8; ++k)
+            dots[qi] = fmaf(kf[k], qg[k], dots[qi]);
+        }
+      }
+#pragma unroll
+      for (int qi = 0; qi < QT; ++qi) {
+        if (qi >= qn) break;
+        // causal mask: q row qi has absolute position pos0 + qi
+        p_s[qi][tid] = (token <= pos0 + qi) ? dots[qi] : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int qi = 0; qi < QT; ++qi) {
+      if (qi >= qn) break;
+      float mine = (tid < clen) ? p_s[qi][tid] : -INFINITY;
+      float cmax = block_max(mine, scratch);
+      if (cmax == -INFINITY) {
+        // fully masked chunk for this row: zero P so phase B adds nothing
+        if (tid < clen) p_s[qi][tid] = 0.f;
+        __syncthreads();
+        continue;
+      }
+      float mn = fmaxf(m[qi], cmax);
+      float alpha = (m[qi] == -INFINITY) ? 0.f : __expf(m[qi] - mn);
+      float p = (tid < clen && p_s[qi][tid] != -INFINITY)
+                    ? __expf(p_s[qi][tid] - mn) : 0.f;
+      if (tid < clen) p_s[qi][tid] = p;
+      float psum = block_sum(p, scratch);
+      l[qi] = l[qi] * alpha + psum;
+      acc[qi] *= alpha;
+      m[qi] = mn;
+    }
+    __syncthreads();
+
+    if (tid < D) {
+      for (int i = 0; i < clen; ++i) {
+        const int token = start + i;
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const bf16 *vrow =
+            vc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+        const float v = bf2f(vrow[tid]);
+#pragma unroll
+        for (int qi = 0; qi < QT; ++qi) {
+          if (qi >= qn) break;
+          acc[qi] = fmaf(p_s[qi][i], v, acc[qi]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (tid < D) {
+#pragma unroll
+    for (int qi = 0; qi < QT; ++qi) {
+      if (qi >= qn) break;
+      const float o = (l[qi] > 0.f) ? acc[qi] / l[qi] : 0.f;
+      out[((long)(q0 + qi) * Hq + h) * D + tid] = f2bf(o);
+    }
+  }
+}

@@ -1,0 +1,204 @@
+// Torch bindings for the quoracle_amd HIP kernels (gfx950).
+//
+// Single translation unit: the kernel .hip files are included directly so no
+// relocatable device code is needed.  Built in-tree by quoracle_amd/ops/build.py
+// with `hipcc -x hip --offload-arch=gfx950` (see __graft_entry__.build()).
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include <stdexcept>
+#include <string>
+
+#include "attention.hip"
+#include "elementwise.hip"
+
+namespace {
+
+#define CHECK_GPU(t)                                                     \
+  TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");                  \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+inline bf16 *bf16_ptr(torch::Tensor &t) {
+  return reinterpret_cast<bf16 *>(t.data_ptr());
+}
+inline const bf16 *bf16_cptr(const torch::Tensor &t) {
+  return reinterpret_cast<const bf16 *>(t.data_ptr());
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void rmsnorm_fused(torch::Tensor y, torch::Tensor x,
+                   c10::optional<torch::Tensor> residual, torch::Tensor w,
+                   double eps) {
+  CHECK_GPU(y);
+  CHECK_GPU(x);
+  CHECK_GPU(w);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+  const int n = x.size(-1);
+  TORCH_CHECK(n % 8 == 0, "hidden size must be a multiple of 8");
+  const long rows = x.numel() / n;
+  bf16 *res_ptr = nullptr;
+  if (residual.has_value()) {
+    CHECK_GPU(residual.value());
+    res_ptr = bf16_ptr(residual.value());
+  }
+  hipLaunchKernelGGL(rmsnorm_fused_kernel, dim3((unsigned)rows), dim3(256), 0,
+                     current_stream(), bf16_ptr(y), bf16_cptr(x), res_ptr,
+                     bf16_cptr(w), n, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
+void swiglu(torch::Tensor out, torch::Tensor gate_up) {
+  CHECK_GPU(out);
+  CHECK_GPU(gate_up);
+  const int inter = out.size(-1);
+  TORCH_CHECK(gate_up.size(-1) == 2 * inter, "gate_up must be [rows, 2*inter]");
+  TORCH_CHECK(inter % 8 == 0, "inter must be a multiple of 8");
+  const long rows = out.numel() / inter;
+  hipLaunchKernelGGL(swiglu_kernel, dim3((unsigned)rows), dim3(256), 0,
+                     current_stream(), bf16_ptr(out), bf16_cptr(gate_up),
+                     inter);
+  HIP_CHECK_KERNEL();
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor pos,
+                  double theta) {
+  CHECK_GPU(q);
+  CHECK_GPU(k);
+  CHECK_GPU(pos);
+  TORCH_CHECK(pos.scalar_type() == torch::kInt32, "pos must be int32");
+  const int T = q.size(0);
+  const int Hq = q.size(1);
+  const int Hk = k.size(1);
+  const int D = q.size(2);
+  TORCH_CHECK(D % 2 == 0 && D <= 512, "bad head dim");
+  if (T == 0) return;
+  hipLaunchKernelGGL(rope_inplace_kernel, dim3(T, Hq + Hk), dim3(D / 2), 0,
+                     current_stream(), bf16_ptr(q), bf16_ptr(k),
+                     pos.data_ptr<int>(), Hq, Hk, D, (float)theta);
+  HIP_CHECK_KERNEL();
+}
+
+void kv_append(torch::Tensor kcache, torch::Tensor vcache, torch::Tensor k,
+               torch::Tensor v, torch::Tensor slots) {
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(k);
+  CHECK_GPU(v);
+  CHECK_GPU(slots);
+  TORCH_CHECK(slots.scalar_type() == torch::kInt32, "slots must be int32");
+  const int T = k.size(0);
+  const int Hk = k.size(1);
+  const int D = k.size(2);
+  const int BS = kcache.size(2);
+  if (T == 0) return;
+  hipLaunchKernelGGL(kv_append_kernel, dim3(T, Hk), dim3(D), 0,
+                     current_stream(), bf16_ptr(kcache), bf16_ptr(vcache),
+                     bf16_cptr(k), bf16_cptr(v), slots.data_ptr<int>(), Hk, BS,
+                     D);
+  HIP_CHECK_KERNEL();
+}
+
+void paged_attn_decode(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor kcache, torch::Tensor vcache,
+                       torch::Tensor block_tables, torch::Tensor ctx_lens,
+                       double scale) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  CHECK_GPU(ctx_lens);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  TORCH_CHECK(Hq % Hkv == 0 && GQ <= MAX_GQ, "unsupported GQA ratio");
+  TORCH_CHECK(D <= DECODE_BLOCK, "head dim too large");
+  if (B == 0) return;
+  hipLaunchKernelGGL(paged_attn_decode_kernel, dim3(B, Hkv),
+                     dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
+                     bf16_cptr(q), bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
+                     (float)scale, Hq, Hkv, D, BS, MAXB, GQ);
+  HIP_CHECK_KERNEL();
+}
+
+void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
+                        torch::Tensor kcache, torch::Tensor vcache,
+                        torch::Tensor block_tables, torch::Tensor tile_q0,
+                        torch::Tensor tile_qn, torch::Tensor tile_seq,
+                        torch::Tensor tile_pos0, double scale) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  const int ntiles = tile_q0.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  if (ntiles == 0) return;
+  hipLaunchKernelGGL(paged_attn_prefill_kernel, dim3(ntiles, Hq),
+                     dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
+                     bf16_cptr(q), bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                     tile_qn.data_ptr<int>(), tile_seq.data_ptr<int>(),
+                     tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, D, BS,
+                     MAXB, GQ);
+  HIP_CHECK_KERNEL();
+}
+
+void cosine_sim_matrix(torch::Tensor out, torch::Tensor x) {
+  CHECK_GPU(out);
+  CHECK_GPU(x);
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32, "x must be f32");
+  const int N = x.size(0);
+  const int D = x.size(1);
+  if (N == 0) return;
+  hipLaunchKernelGGL(cosine_sim_kernel, dim3(N, N), dim3(256), 0,
+                     current_stream(), out.data_ptr<float>(),
+                     x.data_ptr<float>(), N, D);
+  HIP_CHECK_KERNEL();
+}
+
+void gather_rows(torch::Tensor out, torch::Tensor src, torch::Tensor rows) {
+  CHECK_GPU(out);
+  CHECK_GPU(src);
+  CHECK_GPU(rows);
+  const int T = rows.size(0);
+  const int n = src.size(1);
+  TORCH_CHECK(n % 8 == 0, "row width must be a multiple of 8");
+  if (T == 0) return;
+  hipLaunchKernelGGL(gather_rows_kernel, dim3(T), dim3(256), 0,
+                     current_stream(), bf16_ptr(out), bf16_cptr(src),
+                     rows.data_ptr<int>(), n);
+  HIP_CHECK_KERNEL();
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fused", &rmsnorm_fused,
+        "Fused residual-add + RMSNorm (bf16, fp32 accum)");
+  m.def("swiglu", &swiglu, "Fused SiLU(gate) * up");
+  m.def("rope_inplace", &rope_inplace, "Rotate-half RoPE in place on q/k");
+  m.def("kv_append", &kv_append, "Scatter K/V rows into the paged cache");
+  m.def("paged_attn_decode", &paged_attn_decode,
+        "Paged-KV GQA decode attention (one token per sequence)");
+  m.def("paged_attn_prefill", &paged_attn_prefill,
+        "Paged-KV causal prefill attention over cached context");
+  m.def("cosine_sim_matrix", &cosine_sim_matrix,
+        "Pairwise cosine similarity matrix (consensus vote)");
+  m.def("gather_rows", &gather_rows, "Embedding row gather (bf16)");
+}
