@@ -1,0 +1,211 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: consensus agent-steps/sec on an N-GPU model pool.
+
+Measures the BASELINE.json metric — consensus agent-steps per second (and
+p50 step latency) with a 3-model Llama-3-8B pool — through the REAL
+pipeline: every agent step runs per-model constrained decoding on the local
+HIP/CDNA4 engine (random-init weights, synthetic prompts), fingerprint
+clustering, winner/merge selection and action execution.
+
+Scaling is weak: each GPU hosts its own 3-model pool shard and a fixed
+number of agents; rank 0 orchestrates, ranks 1..N-1 serve their models over
+the gloo control plane (one process per GPU, torchrun).
+
+One bench "step" is a fleet round: every agent completes exactly one
+consensus agent-step (all pool models decode, votes clustered, action
+executed).  value = agents * steps / elapsed = whole-job agent-steps/sec.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=4)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--pool-size", type=int, default=3)
+    p.add_argument("--agents-per-gpu", type=int, default=4)
+    p.add_argument("--kv-gb", type=float, default=4.0)
+    p.add_argument("--device", default=None, help="override (e.g. cpu)")
+    return p.parse_args()
+
+
+def rank_model_keys(model: str, pool_size: int, rank: int):
+    return [f"{model}#r{rank}m{j}" for j in range(pool_size)]
+
+
+def main():
+    args = parse_args()
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device(args.device or "cpu")
+
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+        backend = "cpu:gloo,cuda:nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from quoracle_amd.engine.engine import LocalEngine
+
+    local_keys = rank_model_keys(args.model, args.pool_size, rank)
+    engine = LocalEngine(
+        local_keys, device=device, kv_gb_per_model=args.kv_gb,
+        embed_model_key="embed-small" if rank == 0 else None)
+
+    if rank != 0:
+        from quoracle_amd.parallel.control import serve_engine
+        serve_engine(engine)
+        return
+
+    result = asyncio.run(orchestrate(args, engine, device, world))
+    print(json.dumps(result), flush=True)
+
+
+async def orchestrate(args, engine, device, world):
+    import torch
+    from quoracle_amd.agent.core import AgentActor, MESSAGE_TYPES
+    from quoracle_amd.agent.state import AgentState
+    from quoracle_amd.agent.supervisor import Supervisor
+    from quoracle_amd.engine.pool import EnginePool
+    from quoracle_amd.governance.profiles import Profile
+    from quoracle_amd.tasks.runtime import RuntimeConfig, TaskRuntime
+    from quoracle_amd.utils import ids
+
+    engine.start()
+    client = None
+    pool = EnginePool(embedder=engine)
+    if world > 1:
+        from quoracle_amd.parallel.control import ControlClient, RemoteEngine
+        client = ControlClient(list(range(1, world)))
+    for r in range(world):
+        for key in rank_model_keys(args.model, args.pool_size, r):
+            if r == 0:
+                pool.assign(key, engine)
+            else:
+                from quoracle_amd.parallel.control import RemoteEngine
+                pool.assign(key, RemoteEngine(r, client))
+
+    runtime = TaskRuntime(engines=pool, config=RuntimeConfig())
+    Supervisor(runtime)
+    for r in range(world):
+        runtime.profiles.put(Profile(
+            name=f"bench-r{r}", description="bench pool shard",
+            model_pool=rank_model_keys(args.model, args.pool_size, r),
+            capability_groups=[], max_refinement_rounds=2))
+
+    # Build the agent fleet: agents_per_gpu per rank-shard, lockstep-driven
+    # (the actor loop is not started — the bench owns cycle timing).
+    actors = []
+    for i in range(args.agents_per_gpu * world):
+        shard = i % world
+        profile = runtime.profiles.resolve(f"bench-r{shard}")
+        state = AgentState(
+            agent_id=ids.agent_id(f"bench{i}"),
+            task_id=f"bench-task-{i}", parent_id=None,
+            profile=f"bench-r{shard}",
+            model_pool=list(profile.model_pool),
+            capability_groups=[],
+            max_refinement_rounds=profile.max_refinement_rounds,
+        )
+        state.init_model_maps()
+        actor = AgentActor(state, runtime)
+        runtime.registry.register(state.agent_id, actor, state.task_id,
+                                  parent_id=None)
+        actor.state.message_queue.append({
+            "type": "user_message",
+            "content": f"Benchmark task {i}: assess the situation, plan the "
+                       f"work, and coordinate results. Iteration seed {i}."})
+        actors.append(actor)
+
+    step_latencies = []
+
+    async def one_round(actor, timed):
+        t0 = time.perf_counter()
+        while not actor.inbox.empty():
+            msg = actor.inbox.get_nowait()
+            t = msg.get("type")
+            if t == "action_result":
+                actor._handle_action_result(msg)
+            elif t in MESSAGE_TYPES:
+                actor.state.message_queue.append(msg)
+        await actor._run_cycle()
+        if timed:
+            step_latencies.append((time.perf_counter() - t0) * 1e3)
+
+    async def fleet_round(timed=False):
+        await asyncio.gather(*[one_round(a, timed) for a in actors])
+        # let dispatched action tasks deliver their results
+        for _ in range(4):
+            await asyncio.sleep(0)
+
+    def barrier():
+        if world > 1:
+            client.barrier_all()
+        elif torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        await fleet_round()
+    barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        await fleet_round(timed=True)
+    barrier()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        elapsed = client.reduce_max_elapsed(elapsed)
+        client.shutdown()
+    engine.stop()
+
+    n_agents = len(actors)
+    agent_steps = n_agents * args.steps
+    decisions = sum(a.steps_completed for a in actors)
+    return {
+        "metric": "consensus agent-steps/sec",
+        "value": round(agent_steps / elapsed, 3),
+        "unit": "agent-steps/sec",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1e3 / args.steps, 2),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "config": {
+            "model": args.model,
+            "pool_size": args.pool_size,
+            "agents_per_gpu": args.agents_per_gpu,
+            "agents_total": n_agents,
+            "parallelism": f"pool-sharded dp{world}",
+            "p50_step_latency_ms": round(
+                statistics.median(step_latencies), 2) if step_latencies else None,
+            "decisions_completed": decisions,
+            "constrained_decoding": True,
+            "max_refinement_rounds": 2,
+        },
+    }
+
+
+if __name__ == "__main__":
+    main()

@@ -344,6 +344,9 @@ class AgentActor:
                 seed=hash((self.state.agent_id, model_key, round_num)) & 0x7FFFFFFF,
                 action_grammar=True,
                 request_id=ids.request_id(),
+                # stable per (agent, model): consecutive cycles reuse the KV
+                # of the unchanged history prefix (engine prefix cache)
+                session_id=f"{self.state.agent_id}:{model_key}",
             )
             result = await engine.generate(request)
             if result.error == "context_overflow":

@@ -35,7 +35,13 @@ class GenerateRequest:
     # Constrained decoding: force output to be a valid action JSON while the
     # model still does full forwards (the sampler masks logits per template).
     action_grammar: bool = False
+    allowed_actions: Optional[List[str]] = None
     request_id: str = ""
+    # Prefix-cache key: generate calls sharing a session_id reuse the KV of
+    # the longest common token prefix (one session per (agent, model) —
+    # the MI355X analogue of the reference's prompt cache,
+    # reference: agent/consensus_handler.ex:126-152).
+    session_id: str = ""
 
 
 @dataclass

@@ -1,0 +1,400 @@
+"""LocalEngine: continuous-batching inference over the HIP/CDNA4 ops.
+
+MI355X-first design (replaces the reference's HTTP model-access layer,
+reference: lib/quoracle/models/model_query.ex):
+
+  * One LocalEngine per GPU hosts one or more models (288 GB HBM3E holds
+    several 8B-class pool members comfortably).
+  * Every pending agent turn across the whole spawn tree lands in one
+    per-model run queue; each engine step runs ONE mixed forward per model:
+    all decode sequences (1 token each) + chunked prefill tokens, exactly
+    the ForwardBatch the model/kernels were designed around.
+  * A sequence is unified prefill/decode: each step advances the KV cache by
+    a chunk; a 1-token chunk takes the specialized decode-attention kernel.
+  * Conversations are prefix-cached Sessions — consecutive consensus cycles
+    of the same (agent, model) only prefill the newly appended history.
+  * The engine loop runs on its own thread so host-side orchestration
+    (asyncio actor runtime) overlaps GPU compute; results complete asyncio
+    futures via call_soon_threadsafe.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import queue
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+from ..models import LlamaModel, get_config
+from .api import Engine, GenerateRequest, GenerateResult, MIN_OUTPUT_TOKENS
+from .kv_cache import BlockManager, OutOfBlocks, Session, SessionCache
+from .sampler import ActionGrammar, Sampler, SamplingParams
+from .tokenizer import ByteTokenizer, EOS
+from ..models.llama import ForwardBatch, KVCache
+
+BLOCK_SIZE = 16
+PREFILL_CHUNK = 2048         # max prefill tokens per model per engine step
+QT = 16                      # prefill tile height (must match attention.hip)
+
+# $/1M tokens (input, output) per preset — budget plumbing needs real-ish
+# numbers (reference records per-call costs: lib/quoracle/costs/recorder.ex)
+_PRICES = {
+    "tiny": (0.01, 0.02), "gpt2s": (0.02, 0.05),
+    "llama3-8b": (0.2, 0.8), "llama3-70b": (0.9, 3.6),
+    "mixtral-8x7b": (0.5, 2.0), "embed-small": (0.01, 0.01),
+}
+
+
+@dataclass
+class _Seq:
+    request: GenerateRequest
+    prompt: List[int]
+    session: Session
+    params: SamplingParams
+    grammar: Optional[ActionGrammar]
+    generator: Optional[torch.Generator]
+    emitted: List[int] = field(default_factory=list)
+    finished: bool = False
+    t_start: float = field(default_factory=time.monotonic)
+    _complete: Optional[Tuple] = None        # (loop, future) | ("sync", event)
+    result: Optional[GenerateResult] = None
+
+    @property
+    def known(self) -> List[int]:
+        return self.prompt + self.emitted
+
+
+class _HostedModel:
+    def __init__(self, key: str, device: torch.device, kv_blocks: int):
+        self.key = key
+        self.cfg = get_config(key)
+        self.model = LlamaModel(key, device)
+        self.kv: KVCache = self.model.new_kv_cache(kv_blocks, BLOCK_SIZE)
+        self.mgr = BlockManager(kv_blocks, BLOCK_SIZE)
+        self.sessions = SessionCache(self.mgr)
+        self.active: List[_Seq] = []
+        self.sampler = Sampler(self.cfg.vocab_size, device)
+
+
+class LocalEngine(Engine):
+    def __init__(self, model_keys: Sequence[str],
+                 device: Optional[torch.device] = None,
+                 kv_gb_per_model: float = 4.0,
+                 embed_model_key: Optional[str] = "embed-small",
+                 kv_blocks_override: Optional[int] = None,
+                 prefill_chunk: int = PREFILL_CHUNK):
+        self.device = device or torch.device(
+            "cuda:0" if torch.cuda.is_available() else "cpu")
+        self.tokenizer = ByteTokenizer()
+        self.prefill_chunk = prefill_chunk
+        self.models: Dict[str, _HostedModel] = {}
+        for key in model_keys:
+            cfg = get_config(key)
+            if kv_blocks_override is not None:
+                blocks = kv_blocks_override
+            else:
+                per_block = cfg.kv_bytes_per_token() * BLOCK_SIZE
+                blocks = max(8, int(kv_gb_per_model * (1 << 30) / per_block))
+            self.models[key] = _HostedModel(key, self.device, blocks)
+        self.embed_model: Optional[LlamaModel] = None
+        if embed_model_key:
+            self.embed_model = LlamaModel(embed_model_key, self.device)
+        self._inbox: "queue.Queue[_Seq]" = queue.Queue()
+        self._gpu_lock = threading.Lock()
+        self._thread: Optional[threading.Thread] = None
+        self._running = False
+        self._wake = threading.Event()
+
+    # -- lifecycle -----------------------------------------------------------
+
+    def start(self) -> "LocalEngine":
+        if self._thread is None:
+            self._running = True
+            self._thread = threading.Thread(target=self._loop, daemon=True,
+                                            name="quoracle-engine")
+            self._thread.start()
+        return self
+
+    def stop(self) -> None:
+        self._running = False
+        self._wake.set()
+        if self._thread:
+            self._thread.join(timeout=30)
+            self._thread = None
+
+    # -- Engine protocol -----------------------------------------------------
+
+    async def generate(self, request: GenerateRequest) -> GenerateResult:
+        loop = asyncio.get_running_loop()
+        fut: asyncio.Future = loop.create_future()
+        err = self._submit(request, ("async", loop, fut))
+        if err is not None:
+            return err
+        return await fut
+
+    def generate_sync(self, request: GenerateRequest,
+                      timeout: Optional[float] = None) -> GenerateResult:
+        """Blocking path for tests / non-async callers; drives the engine
+        inline when the loop thread is not running."""
+        ev = threading.Event()
+        box: List[GenerateResult] = []
+        err = self._submit(request, ("sync", ev, box))
+        if err is not None:
+            return err
+        if self._thread is None:
+            deadline = None if timeout is None else time.monotonic() + timeout
+            while not ev.is_set():
+                self.step()
+                if deadline and time.monotonic() > deadline:
+                    raise TimeoutError("generate_sync timed out")
+        else:
+            ev.wait(timeout)
+        return box[0]
+
+    async def embed(self, texts: List[str]) -> Sequence[Sequence[float]]:
+        return await asyncio.get_running_loop().run_in_executor(
+            None, self.embed_sync, list(texts))
+
+    def embed_sync(self, texts: List[str]) -> Sequence[Sequence[float]]:
+        if self.embed_model is None:
+            raise RuntimeError("no embedding model hosted")
+        batches = []
+        for t in texts:
+            ids = self.tokenizer.encode(t)[:512] or [EOS]
+            batches.append(torch.tensor(ids, dtype=torch.int32,
+                                        device=self.device))
+        with self._gpu_lock:
+            hid = self.embed_model.embed_texts_hidden(batches)
+        norm = hid.norm(dim=-1, keepdim=True).clamp_min(1e-8)
+        return (hid / norm).cpu().tolist()
+
+    def count_tokens(self, text: str) -> int:
+        return self.tokenizer.count(text)
+
+    def context_limit(self, model_key: str) -> int:
+        return get_config(model_key).max_context
+
+    def output_limit(self, model_key: str) -> int:
+        return get_config(model_key).max_output
+
+    def drop_session(self, model_key: str, session_id: str) -> None:
+        hm = self.models.get(model_key)
+        if hm:
+            hm.sessions.drop(session_id)
+
+    # -- request intake ------------------------------------------------------
+
+    def _submit(self, request: GenerateRequest, completion) -> Optional[GenerateResult]:
+        hm = self.models.get(request.model_key)
+        if hm is None:
+            return GenerateResult(model_key=request.model_key,
+                                  error=f"model_not_hosted:{request.model_key}")
+        prompt = self.tokenizer.render_messages(request.messages)
+        if len(prompt) > hm.cfg.max_context:
+            return GenerateResult(model_key=request.model_key,
+                                  error="context_overflow",
+                                  input_tokens=len(prompt))
+        params = SamplingParams(
+            temperature=request.temperature, top_p=request.top_p,
+            max_tokens=min(request.max_tokens, hm.cfg.max_output,
+                           hm.cfg.max_context - len(prompt)),
+            seed=request.seed)
+        grammar = None
+        if request.action_grammar:
+            grammar = ActionGrammar(request.allowed_actions
+                                    or ["orient", "send_message", "todo", "wait"])
+        gen = None
+        if request.seed is not None:
+            gen = torch.Generator(device=self.device)
+            gen.manual_seed(request.seed)
+        sid = request.session_id or f"anon-{id(request)}"
+        seq = _Seq(request=request, prompt=prompt, session=None,  # type: ignore
+                   params=params, grammar=grammar, generator=gen,
+                   _complete=completion)
+        seq._session_id = sid                                     # type: ignore
+        self._inbox.put(seq)
+        self._wake.set()
+        return None
+
+    # -- engine loop ---------------------------------------------------------
+
+    def _loop(self) -> None:
+        while self._running:
+            worked = self.step()
+            if not worked:
+                self._wake.wait(timeout=0.005)
+                self._wake.clear()
+
+    def step(self) -> bool:
+        """One engine step: admit new work, run one mixed forward per model
+        with pending sequences.  Returns True if any work was done."""
+        self._admit()
+        worked = False
+        for hm in self.models.values():
+            if hm.active:
+                with self._gpu_lock:
+                    self._step_model(hm)
+                worked = True
+        return worked
+
+    def _admit(self) -> None:
+        while True:
+            try:
+                seq = self._inbox.get_nowait()
+            except queue.Empty:
+                return
+            hm = self.models[seq.request.model_key]
+            sess = hm.sessions.get_or_create(seq._session_id)   # type: ignore
+            hm.sessions.match_prefix(sess, seq.prompt)
+            seq.session = sess
+            hm.active.append(seq)
+
+    def _step_model(self, hm: _HostedModel) -> None:
+        dev = self.device
+        decode: List[_Seq] = []
+        prefill: List[Tuple[_Seq, int]] = []       # (seq, n_new_tokens)
+        budget = self.prefill_chunk
+        for seq in hm.active:
+            cached = len(seq.session.token_ids)
+            remaining = len(seq.known) - cached
+            if remaining <= 0:
+                remaining = 1      # shouldn't happen; treat as decode resample
+            if remaining == 1:
+                decode.append(seq)
+            elif budget > 0:
+                n = min(remaining, budget)
+                budget -= n
+                prefill.append((seq, n))
+
+        if not decode and not prefill:
+            return
+
+        tokens: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        bt_rows: List[List[int]] = []
+        ctx_lens: List[int] = []
+        failed: List[_Seq] = []
+        active_ids = [s.session.session_id for s in hm.active]
+
+        def _extend(seq: _Seq, toks: List[int]) -> Optional[List[int]]:
+            try:
+                return hm.sessions.extend(seq.session, toks, active=active_ids)
+            except OutOfBlocks as exc:
+                seq.result = GenerateResult(model_key=hm.key,
+                                            error=f"kv_exhausted:{exc}")
+                failed.append(seq)
+                return None
+
+        for seq in decode:
+            cached = len(seq.session.token_ids)
+            tok = seq.known[cached]
+            s = _extend(seq, [tok])
+            if s is None:
+                continue
+            tokens.append(tok)
+            positions.append(cached)
+            slots.extend(s)
+            bt_rows.append(list(seq.session.blocks))
+            ctx_lens.append(cached + 1)
+        n_decode = len(ctx_lens)
+
+        tile_q0: List[int] = []
+        tile_qn: List[int] = []
+        tile_seq: List[int] = []
+        tile_pos0: List[int] = []
+        # decode rows sample at their own row index
+        sample_rows: List[int] = list(range(n_decode))
+        sample_seqs: List[_Seq] = [s for s in decode if s not in failed]
+
+        row = n_decode
+        for seq, n in prefill:
+            cached = len(seq.session.token_ids)
+            chunk = seq.known[cached:cached + n]
+            s = _extend(seq, chunk)
+            if s is None:
+                continue
+            bt_rows.append(list(seq.session.blocks))
+            seq_row = len(bt_rows) - 1
+            tokens.extend(chunk)
+            positions.extend(range(cached, cached + n))
+            slots.extend(s)
+            for off in range(0, n, QT):
+                tile_q0.append(row + off)
+                tile_qn.append(min(QT, n - off))
+                tile_seq.append(seq_row)
+                tile_pos0.append(cached + off)
+            if cached + n == len(seq.known):
+                sample_rows.append(row + n - 1)
+                sample_seqs.append(seq)
+            row += n
+
+        if not tokens:
+            self._finish(hm, failed)
+            return
+
+        maxb = max(len(r) for r in bt_rows)
+        bt = torch.zeros((len(bt_rows), maxb), dtype=torch.int32)
+        for i, r in enumerate(bt_rows):
+            bt[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+
+        def t32(x):
+            return torch.tensor(x, dtype=torch.int32, device=dev)
+
+        batch = ForwardBatch(
+            tokens=t32(tokens), positions=t32(positions), slots=t32(slots),
+            block_tables=bt.to(dev), n_decode=n_decode,
+            ctx_lens=t32(ctx_lens) if n_decode else None,
+            tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
+            tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
+
+        hidden = hm.model.forward(batch, hm.kv)
+        done: List[_Seq] = list(failed)
+        if sample_rows:
+            rows = torch.tensor(sample_rows, dtype=torch.long, device=dev)
+            logits = hm.model.compute_logits(hidden, rows)
+            ids = hm.sampler.sample(
+                logits, [s.params for s in sample_seqs],
+                [s.grammar for s in sample_seqs],
+                [s.generator for s in sample_seqs])
+            for seq, tok in zip(sample_seqs, ids):
+                seq.emitted.append(tok)
+                if (tok == EOS or len(seq.emitted) >= seq.params.max_tokens
+                        or (seq.grammar is not None and seq.grammar.done)):
+                    seq.finished = True
+                    seq.result = self._make_result(hm, seq)
+                    done.append(seq)
+        self._finish(hm, done)
+
+    def _make_result(self, hm: _HostedModel, seq: _Seq) -> GenerateResult:
+        out_ids = [t for t in seq.emitted if t != EOS]
+        text = self.tokenizer.decode(out_ids)
+        n_in, n_out = len(seq.prompt), len(seq.emitted)
+        pin, pout = _PRICES.get(hm.cfg.name, (0.1, 0.4))
+        return GenerateResult(
+            model_key=hm.key, text=text, input_tokens=n_in,
+            output_tokens=n_out,
+            latency_ms=(time.monotonic() - seq.t_start) * 1e3,
+            cost=(n_in * pin + n_out * pout) / 1e6)
+
+    def _finish(self, hm: _HostedModel, done: List[_Seq]) -> None:
+        for seq in done:
+            if seq in hm.active:
+                hm.active.remove(seq)
+            comp = seq._complete
+            if comp is None:
+                continue
+            if comp[0] == "async":
+                _, loop, fut = comp
+                loop.call_soon_threadsafe(
+                    lambda f=fut, r=seq.result: f.done() or f.set_result(r))
+            elif comp[0] == "cb":
+                comp[1](seq.result)
+            else:
+                _, ev, box = comp
+                box.append(seq.result)
+                ev.set()

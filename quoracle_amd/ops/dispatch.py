@@ -1,0 +1,90 @@
+"""Device dispatch for the compute ops.
+
+GPU tensors go to the HIP extension and FAIL LOUDLY if it is missing — a
+silent eager fallback can never masquerade as the native path on a GPU box.
+CPU tensors go to the fp32 reference implementations so the whole engine
+(model forward, KV pager, batcher) is testable without a GPU.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import ext
+from . import reference as ref
+
+
+def rmsnorm(y: torch.Tensor, x: torch.Tensor,
+            residual: Optional[torch.Tensor], w: torch.Tensor,
+            eps: float = 1e-5) -> torch.Tensor:
+    """y = rmsnorm(x (+ residual)) * w; residual updated in place on GPU."""
+    if x.is_cuda:
+        ext().rmsnorm_fused(y, x, residual, w, eps)
+        return y
+    out, new_res = ref.rmsnorm(x, w, residual=residual, eps=eps)
+    y.copy_(out.to(y.dtype))
+    if residual is not None:
+        residual.copy_(new_res.to(residual.dtype))
+    return y
+
+
+def swiglu(out: torch.Tensor, gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        ext().swiglu(out, gate_up)
+        return out
+    out.copy_(ref.swiglu(gate_up).to(out.dtype))
+    return out
+
+
+def rope_inplace(q: torch.Tensor, k: torch.Tensor, pos: torch.Tensor,
+                 theta: float) -> None:
+    if q.is_cuda:
+        ext().rope_inplace(q, k, pos, theta)
+        return
+    q.copy_(ref.rope(q, pos, theta=theta).to(q.dtype))
+    k.copy_(ref.rope(k, pos, theta=theta).to(k.dtype))
+
+
+def kv_append(kcache, vcache, k, v, slots) -> None:
+    if kcache.is_cuda:
+        ext().kv_append(kcache, vcache, k, v, slots)
+        return
+    ref.kv_append(kcache, vcache, k, v, slots)
+
+
+def paged_attn_decode(out, q, kcache, vcache, block_tables, ctx_lens,
+                      scale: float):
+    if q.is_cuda:
+        ext().paged_attn_decode(out, q, kcache, vcache, block_tables,
+                                ctx_lens, scale)
+        return out
+    return ref.paged_attn_decode(out, q, kcache, vcache, block_tables,
+                                 ctx_lens, scale)
+
+
+def paged_attn_prefill(out, q, kcache, vcache, block_tables, tile_q0,
+                       tile_qn, tile_seq, tile_pos0, scale: float):
+    if q.is_cuda:
+        ext().paged_attn_prefill(out, q, kcache, vcache, block_tables,
+                                 tile_q0, tile_qn, tile_seq, tile_pos0, scale)
+        return out
+    return ref.paged_attn_prefill(out, q, kcache, vcache, block_tables,
+                                  tile_q0, tile_qn, tile_seq, tile_pos0,
+                                  scale)
+
+
+def gather_rows(out, src, rows):
+    if src.is_cuda:
+        ext().gather_rows(out, src, rows)
+        return out
+    return ref.gather_rows(out, src, rows)
+
+
+def cosine_sim_matrix(out, x):
+    if x.is_cuda:
+        ext().cosine_sim_matrix(out, x)
+        return out
+    out.copy_(ref.cosine_sim_matrix(x))
+    return out

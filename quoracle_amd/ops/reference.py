@@ -81,3 +81,65 @@ def cosine_sim_matrix(x: torch.Tensor) -> torch.Tensor:
     sim[zero, :] = 0.0
     sim[:, zero] = 0.0
     return sim
+
+
+# -- paged-KV reference implementations (same layouts as the HIP kernels) ----
+# kcache/vcache: [NB, Hkv, BS, D]; block_tables: [B, MAXB] int32;
+# slots: [T] int32 global slot = block*BS + offset.
+
+def kv_append(kcache: torch.Tensor, vcache: torch.Tensor,
+              k: torch.Tensor, v: torch.Tensor, slots: torch.Tensor) -> None:
+    NB, Hkv, BS, D = kcache.shape
+    for t in range(slots.shape[0]):
+        s = int(slots[t])
+        if s < 0:
+            continue
+        blk, off = s // BS, s % BS
+        kcache[blk, :, off, :] = k[t]
+        vcache[blk, :, off, :] = v[t]
+
+
+def _gather_kv(cache: torch.Tensor, block_table: torch.Tensor,
+               length: int) -> torch.Tensor:
+    """[len, Hkv, D] rows 0..len-1 of one sequence from the paged cache."""
+    NB, Hkv, BS, D = cache.shape
+    toks = torch.arange(length, device=cache.device)
+    blocks = block_table[toks // BS].long()
+    offs = toks % BS
+    return cache[blocks, :, offs, :]          # [len, Hkv, D]
+
+
+def paged_attn_decode(out: torch.Tensor, q: torch.Tensor,
+                      kcache: torch.Tensor, vcache: torch.Tensor,
+                      block_tables: torch.Tensor, ctx_lens: torch.Tensor,
+                      scale: float) -> torch.Tensor:
+    B = q.shape[0]
+    for b in range(B):
+        length = int(ctx_lens[b])
+        k = _gather_kv(kcache, block_tables[b], length)
+        v = _gather_kv(vcache, block_tables[b], length)
+        out[b] = attention(q[b:b + 1].float(), k, v, scale)[0].to(out.dtype)
+    return out
+
+
+def paged_attn_prefill(out: torch.Tensor, q: torch.Tensor,
+                       kcache: torch.Tensor, vcache: torch.Tensor,
+                       block_tables: torch.Tensor,
+                       tile_q0: torch.Tensor, tile_qn: torch.Tensor,
+                       tile_seq: torch.Tensor, tile_pos0: torch.Tensor,
+                       scale: float) -> torch.Tensor:
+    for t in range(tile_q0.shape[0]):
+        q0, qn = int(tile_q0[t]), int(tile_qn[t])
+        seq, pos0 = int(tile_seq[t]), int(tile_pos0[t])
+        kv_len = pos0 + qn
+        k = _gather_kv(kcache, block_tables[seq], kv_len)
+        v = _gather_kv(vcache, block_tables[seq], kv_len)
+        o = attention(q[q0:q0 + qn].float(), k, v, scale, causal_offset=pos0)
+        out[q0:q0 + qn] = o.to(out.dtype)
+    return out
+
+
+def gather_rows(out: torch.Tensor, src: torch.Tensor,
+                rows: torch.Tensor) -> torch.Tensor:
+    out.copy_(src[rows.long()])
+    return out
