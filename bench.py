@@ -44,9 +44,14 @@ def rank_model_keys(model: str, pool_size: int, rank: int):
     return [f"{model}#r{rank}m{j}" for j in range(pool_size)]
 
 
+T_START = time.perf_counter()
+
+
 def main():
     args = parse_args()
     import torch
+    print(f"[bench t={time.perf_counter() - T_START:.1f}s] torch imported",
+          file=sys.stderr, flush=True)
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -70,6 +75,9 @@ def main():
     engine = LocalEngine(
         local_keys, device=device, kv_gb_per_model=args.kv_gb,
         embed_model_key="embed-small" if rank == 0 else None)
+    print(f"[bench t={time.perf_counter() - T_START:.1f}s] rank {rank}: "
+          f"engine up, models {local_keys} on {device}",
+          file=sys.stderr, flush=True)
 
     if rank != 0:
         from quoracle_amd.parallel.control import serve_engine
@@ -163,14 +171,25 @@ async def orchestrate(args, engine, device, world):
         elif torch.cuda.is_available():
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    def note(msg):
+        print(f"[bench t={time.perf_counter() - T_START:.1f}s] {msg}",
+              file=sys.stderr, flush=True)
+
+    note(f"agents built: {len(actors)}; engine stats {engine.stats}")
+    for i in range(args.warmup):
+        tw = time.perf_counter()
         await fleet_round()
+        note(f"warmup round {i}: {time.perf_counter() - tw:.2f}s "
+             f"stats={engine.stats}")
     barrier()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
+        tw = time.perf_counter()
         await fleet_round(timed=True)
+        note(f"timed round {i}: {time.perf_counter() - tw:.2f}s")
     barrier()
     elapsed = time.perf_counter() - t0
+    note(f"timed region done: {elapsed:.2f}s stats={engine.stats}")
     if world > 1:
         elapsed = client.reduce_max_elapsed(elapsed)
         client.shutdown()

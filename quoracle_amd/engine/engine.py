@@ -21,7 +21,9 @@ reference: lib/quoracle/models/model_query.ex):
 from __future__ import annotations
 
 import asyncio
+import os
 import queue
+import sys
 import threading
 import time
 from dataclasses import dataclass, field
@@ -108,6 +110,11 @@ class LocalEngine(Engine):
         self._thread: Optional[threading.Thread] = None
         self._running = False
         self._wake = threading.Event()
+        # telemetry (read by bench/monitor; written only by the engine thread)
+        self.stats = {"engine_steps": 0, "forward_tokens": 0,
+                      "decode_tokens": 0, "prefill_tokens": 0,
+                      "requests_done": 0, "prefix_hit_tokens": 0}
+        self._log_every = int(os.environ.get("QUORACLE_ENGINE_LOG", "0"))
 
     # -- lifecycle -----------------------------------------------------------
 
@@ -198,11 +205,15 @@ class LocalEngine(Engine):
             return GenerateResult(model_key=request.model_key,
                                   error="context_overflow",
                                   input_tokens=len(prompt))
+        max_tokens = min(request.max_tokens, hm.cfg.max_output,
+                         hm.cfg.max_context - len(prompt))
+        if request.action_grammar:
+            # the grammar terminates by itself (~600 tokens worst case);
+            # never truncate it into unparseable JSON
+            max_tokens = max(max_tokens, 1024)
         params = SamplingParams(
             temperature=request.temperature, top_p=request.top_p,
-            max_tokens=min(request.max_tokens, hm.cfg.max_output,
-                           hm.cfg.max_context - len(prompt)),
-            seed=request.seed)
+            max_tokens=max_tokens, seed=request.seed)
         grammar = None
         if request.action_grammar:
             grammar = ActionGrammar(request.allowed_actions
@@ -236,8 +247,20 @@ class LocalEngine(Engine):
         worked = False
         for hm in self.models.values():
             if hm.active:
-                with self._gpu_lock:
-                    self._step_model(hm)
+                try:
+                    with self._gpu_lock:
+                        self._step_model(hm)
+                except Exception as exc:  # noqa: BLE001 — fail loud, not hang
+                    import traceback
+                    traceback.print_exc()
+                    print(f"[engine {hm.key}] step failed: {exc}; failing "
+                          f"{len(hm.active)} in-flight requests",
+                          file=sys.stderr, flush=True)
+                    crashed = list(hm.active)
+                    for seq in crashed:
+                        seq.result = GenerateResult(
+                            model_key=hm.key, error=f"engine_error:{exc}")
+                    self._finish(hm, crashed)
                 worked = True
         return worked
 
@@ -249,7 +272,8 @@ class LocalEngine(Engine):
                 return
             hm = self.models[seq.request.model_key]
             sess = hm.sessions.get_or_create(seq._session_id)   # type: ignore
-            hm.sessions.match_prefix(sess, seq.prompt)
+            hit = hm.sessions.match_prefix(sess, seq.prompt)
+            self.stats["prefix_hit_tokens"] += hit
             seq.session = sess
             hm.active.append(seq)
 
@@ -352,7 +376,18 @@ class LocalEngine(Engine):
             tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
             tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
 
+        t_fwd = time.monotonic()
         hidden = hm.model.forward(batch, hm.kv)
+        st = self.stats
+        st["engine_steps"] += 1
+        st["forward_tokens"] += len(tokens)
+        st["decode_tokens"] += n_decode
+        st["prefill_tokens"] += len(tokens) - n_decode
+        if self._log_every and st["engine_steps"] % self._log_every == 0:
+            print(f"[engine {hm.key}] step={st['engine_steps']} "
+                  f"T={len(tokens)} dec={n_decode} active={len(hm.active)} "
+                  f"fwd_build_ms={(time.monotonic() - t_fwd) * 1e3:.1f}",
+                  file=sys.stderr, flush=True)
         done: List[_Seq] = list(failed)
         if sample_rows:
             rows = torch.tensor(sample_rows, dtype=torch.long, device=dev)
@@ -367,6 +402,7 @@ class LocalEngine(Engine):
                         or (seq.grammar is not None and seq.grammar.done)):
                     seq.finished = True
                     seq.result = self._make_result(hm, seq)
+                    st["requests_done"] += 1
                     done.append(seq)
         self._finish(hm, done)
 

@@ -17,7 +17,7 @@ def run_smoke(model_key: str = "llama3-8b") -> dict:
         model_key=model_key,
         messages=[{"role": "system", "content": "You are a consensus agent."},
                   {"role": "user", "content": "Assess the task."}],
-        temperature=0.8, max_tokens=256, seed=7,
+        temperature=0.8, max_tokens=1024, seed=7,
         action_grammar=True, session_id="smoke")
     result = engine.generate_sync(req, timeout=600)
     assert result.ok, f"smoke generate failed: {result.error}"
