@@ -80,6 +80,15 @@ class _HostedModel:
         self.sessions = SessionCache(self.mgr)
         self.active: List[_Seq] = []
         self.sampler = Sampler(self.cfg.vocab_size, device)
+        # each hosted model launches on its own HIP stream so pool members
+        # overlap on the GPU instead of serializing through the engine loop
+        self.stream = (torch.cuda.Stream(device)
+                       if device.type == "cuda" else None)
+
+    def stream_ctx(self):
+        import contextlib
+        return (torch.cuda.stream(self.stream) if self.stream is not None
+                else contextlib.nullcontext())
 
 
 class LocalEngine(Engine):
@@ -120,6 +129,8 @@ class LocalEngine(Engine):
 
     def start(self) -> "LocalEngine":
         if self._thread is None:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize(self.device)   # weight init done
             self._running = True
             self._thread = threading.Thread(target=self._loop, daemon=True,
                                             name="quoracle-engine")
@@ -241,28 +252,42 @@ class LocalEngine(Engine):
                 self._wake.clear()
 
     def step(self) -> bool:
-        """One engine step: admit new work, run one mixed forward per model
-        with pending sequences.  Returns True if any work was done."""
+        """One engine step: admit new work, then for every model with pending
+        sequences LAUNCH one mixed forward on that model's own stream (phase
+        1, async) and only then sample (phase 2, syncs per stream) — pool
+        members overlap on the GPU.  Returns True if any work was done."""
         self._admit()
-        worked = False
+        pending = []
         for hm in self.models.values():
-            if hm.active:
-                try:
-                    with self._gpu_lock:
-                        self._step_model(hm)
-                except Exception as exc:  # noqa: BLE001 — fail loud, not hang
-                    import traceback
-                    traceback.print_exc()
-                    print(f"[engine {hm.key}] step failed: {exc}; failing "
-                          f"{len(hm.active)} in-flight requests",
-                          file=sys.stderr, flush=True)
-                    crashed = list(hm.active)
-                    for seq in crashed:
-                        seq.result = GenerateResult(
-                            model_key=hm.key, error=f"engine_error:{exc}")
-                    self._finish(hm, crashed)
-                worked = True
-        return worked
+            if not hm.active:
+                continue
+            try:
+                with self._gpu_lock:
+                    ctxt = self._launch_model(hm)
+            except Exception as exc:  # noqa: BLE001 — fail loud, not hang
+                self._crash_model(hm, exc)
+                continue
+            if ctxt is not None:
+                pending.append((hm, ctxt))
+        for hm, ctxt in pending:
+            try:
+                with self._gpu_lock:
+                    self._sample_model(hm, *ctxt)
+            except Exception as exc:  # noqa: BLE001
+                self._crash_model(hm, exc)
+        return bool(pending)
+
+    def _crash_model(self, hm: _HostedModel, exc: Exception) -> None:
+        import traceback
+        traceback.print_exc()
+        print(f"[engine {hm.key}] step failed: {exc}; failing "
+              f"{len(hm.active)} in-flight requests",
+              file=sys.stderr, flush=True)
+        crashed = list(hm.active)
+        for seq in crashed:
+            seq.result = GenerateResult(
+                model_key=hm.key, error=f"engine_error:{exc}")
+        self._finish(hm, crashed)
 
     def _admit(self) -> None:
         while True:
@@ -277,7 +302,7 @@ class LocalEngine(Engine):
             seq.session = sess
             hm.active.append(seq)
 
-    def _step_model(self, hm: _HostedModel) -> None:
+    def _launch_model(self, hm: _HostedModel):
         dev = self.device
         decode: List[_Seq] = []
         prefill: List[Tuple[_Seq, int]] = []       # (seq, n_new_tokens)
@@ -295,7 +320,7 @@ class LocalEngine(Engine):
                 prefill.append((seq, n))
 
         if not decode and not prefill:
-            return
+            return None
 
         tokens: List[int] = []
         positions: List[int] = []
@@ -359,7 +384,7 @@ class LocalEngine(Engine):
 
         if not tokens:
             self._finish(hm, failed)
-            return
+            return None
 
         maxb = max(len(r) for r in bt_rows)
         bt = torch.zeros((len(bt_rows), maxb), dtype=torch.int32)
@@ -369,15 +394,23 @@ class LocalEngine(Engine):
         def t32(x):
             return torch.tensor(x, dtype=torch.int32, device=dev)
 
-        batch = ForwardBatch(
-            tokens=t32(tokens), positions=t32(positions), slots=t32(slots),
-            block_tables=bt.to(dev), n_decode=n_decode,
-            ctx_lens=t32(ctx_lens) if n_decode else None,
-            tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
-            tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
-
         t_fwd = time.monotonic()
-        hidden = hm.model.forward(batch, hm.kv)
+        # batch tensors are H2D copies: build them on the model's stream so
+        # the forward (same stream) is ordered after them without a sync
+        with hm.stream_ctx():
+            batch = ForwardBatch(
+                tokens=t32(tokens), positions=t32(positions), slots=t32(slots),
+                block_tables=bt.to(dev, non_blocking=False),
+                n_decode=n_decode,
+                ctx_lens=t32(ctx_lens) if n_decode else None,
+                max_ctx=max(ctx_lens) if ctx_lens else 0,
+                tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
+                tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
+            hidden = hm.model.forward(batch, hm.kv)
+            logits = None
+            if sample_rows:
+                rows = torch.tensor(sample_rows, dtype=torch.long, device=dev)
+                logits = hm.model.compute_logits(hidden, rows)
         st = self.stats
         st["engine_steps"] += 1
         st["forward_tokens"] += len(tokens)
@@ -386,23 +419,26 @@ class LocalEngine(Engine):
         if self._log_every and st["engine_steps"] % self._log_every == 0:
             print(f"[engine {hm.key}] step={st['engine_steps']} "
                   f"T={len(tokens)} dec={n_decode} active={len(hm.active)} "
-                  f"fwd_build_ms={(time.monotonic() - t_fwd) * 1e3:.1f}",
+                  f"launch_ms={(time.monotonic() - t_fwd) * 1e3:.1f}",
                   file=sys.stderr, flush=True)
+        return (sample_seqs, logits, failed)
+
+    def _sample_model(self, hm: _HostedModel, sample_seqs, logits,
+                      failed) -> None:
         done: List[_Seq] = list(failed)
-        if sample_rows:
-            rows = torch.tensor(sample_rows, dtype=torch.long, device=dev)
-            logits = hm.model.compute_logits(hidden, rows)
-            ids = hm.sampler.sample(
-                logits, [s.params for s in sample_seqs],
-                [s.grammar for s in sample_seqs],
-                [s.generator for s in sample_seqs])
+        if logits is not None:
+            with hm.stream_ctx():
+                ids = hm.sampler.sample(
+                    logits, [s.params for s in sample_seqs],
+                    [s.grammar for s in sample_seqs],
+                    [s.generator for s in sample_seqs])
             for seq, tok in zip(sample_seqs, ids):
                 seq.emitted.append(tok)
                 if (tok == EOS or len(seq.emitted) >= seq.params.max_tokens
                         or (seq.grammar is not None and seq.grammar.done)):
                     seq.finished = True
                     seq.result = self._make_result(hm, seq)
-                    st["requests_done"] += 1
+                    self.stats["requests_done"] += 1
                     done.append(seq)
         self._finish(hm, done)
 

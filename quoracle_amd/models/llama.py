@@ -44,6 +44,7 @@ class ForwardBatch:
     block_tables: torch.Tensor    # [B_all, MAXB] int32 (decode seqs first)
     n_decode: int = 0
     ctx_lens: Optional[torch.Tensor] = None   # [n_decode] int32 (incl. new tok)
+    max_ctx: int = 0                          # host-known max(ctx_lens)
     # prefill tiles (QT=16 query rows each); empty tensors when no prefill
     tile_q0: Optional[torch.Tensor] = None    # [ntiles] int32 row into tokens
     tile_qn: Optional[torch.Tensor] = None
@@ -165,7 +166,7 @@ class LlamaModel:
                 D.paged_attn_decode(
                     attn_out[:batch.n_decode], q[:batch.n_decode],
                     kv.k[li], kv.v[li], batch.block_tables,
-                    batch.ctx_lens, self.scale)
+                    batch.ctx_lens, self.scale, max_ctx=batch.max_ctx)
             if has_prefill:
                 D.paged_attn_prefill(
                     attn_out, q, kv.k[li], kv.v[li], batch.block_tables,

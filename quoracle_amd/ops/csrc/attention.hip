@@ -281,3 +281,168 @@ paged_attn_prefill_kernel(bf16 *__restrict__ out, const bf16 *__restrict__ q,
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Flash-decoding split kernels: the single-pass decode kernel launches only
+// B*Hkv workgroups — at agent-pool batch sizes (B<=16) that is <6% of the
+// 256 CUs and the context walk serializes.  Split the context across NSPLIT
+// workgroups per (seq, kv-head), each producing partial online-softmax state
+// (m, l, acc[D]) to a workspace, then combine per query head.
+//
+//   part_m/part_l: [B, Hq, NS] f32
+//   part_acc:      [B, Hq, NS, D] f32
+// Grid: (B, Hkv, NS); block: DECODE_BLOCK.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_decode_split_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ ctx, float scale,
+    int Hq, int Hkv, int D, int BS, int MAXB, int GQ, int NS) {
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int split = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int len = ctx[b];
+  // contiguous token range for this split (CHUNK-aligned)
+  const int chunks = (len + CHUNK - 1) / CHUNK;
+  const int per = (chunks + NS - 1) / NS;
+  const int t0 = split * per * CHUNK;
+  const int t1 = min(len, (split + 1) * per * CHUNK);
+  const bool dead = (t0 >= len);
+
+  __shared__ float q_s[MAX_GQ * 128];
+  __shared__ float p_s[MAX_GQ][CHUNK];
+  __shared__ float scratch[8];
+
+  for (int i = tid; i < GQ * D; i += blockDim.x) {
+    int g = i / D, d = i % D;
+    q_s[g * D + d] = bf2f(q[((long)b * Hq + hk * GQ + g) * D + d]) * scale;
+  }
+  __syncthreads();
+
+  float m[MAX_GQ], l[MAX_GQ], acc[MAX_GQ];
+#pragma unroll
+  for (int g = 0; g < MAX_GQ; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+    acc[g] = 0.f;
+  }
+
+  const long panel_stride = (long)Hkv * BS * D;
+  for (int start = t0; start < t1; start += CHUNK) {
+    const int clen = min(CHUNK, t1 - start);
+    if (tid < clen) {
+      const int token = start + tid;
+      const long blk = bt[(long)b * MAXB + token / BS];
+      const bf16 *krow =
+          kc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+      float dots[MAX_GQ];
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) dots[g] = 0.f;
+      for (int d8 = 0; d8 < D / 8; ++d8) {
+        uint4 kv = reinterpret_cast<const uint4 *>(krow)[d8];
+        float kf[8];
+        unpack_bf16x2(kv.x, kf[0], kf[1]);
+        unpack_bf16x2(kv.y, kf[2], kf[3]);
+        unpack_bf16x2(kv.z, kf[4], kf[5]);
+        unpack_bf16x2(kv.w, kf[6], kf[7]);
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          const float *qg = q_s + g * D + d8 * 8;
+#pragma unroll
+          for (int k = 0; k < 8; ++k) dots[g] = fmaf(kf[k], qg[k], dots[g]);
+        }
+      }
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) {
+        if (g >= GQ) break;
+        p_s[g][tid] = dots[g];
+      }
+    }
+    __syncthreads();
+
+    for (int g = 0; g < GQ; ++g) {
+      float mine = (tid < clen) ? p_s[g][tid] : -INFINITY;
+      float cmax = block_max(mine, scratch);
+      float mn = fmaxf(m[g], cmax);
+      float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+      float p = (tid < clen) ? __expf(p_s[g][tid] - mn) : 0.f;
+      if (tid < clen) p_s[g][tid] = p;
+      float psum = block_sum(p, scratch);
+      l[g] = l[g] * alpha + psum;
+      acc[g] *= alpha;
+      m[g] = mn;
+    }
+    __syncthreads();
+
+    if (tid < D) {
+      for (int i = 0; i < clen; ++i) {
+        const int token = start + i;
+        const long blk = bt[(long)b * MAXB + token / BS];
+        const bf16 *vrow =
+            vc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+        const float v = bf2f(vrow[tid]);
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          acc[g] = fmaf(p_s[g][i], v, acc[g]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (tid < D) {
+#pragma unroll
+    for (int g = 0; g < MAX_GQ; ++g) {
+      if (g >= GQ) break;
+      const int h = hk * GQ + g;
+      const long base = ((long)b * Hq + h) * NS + split;
+      if (tid == 0) {
+        part_m[base] = dead ? -INFINITY : m[g];
+        part_l[base] = dead ? 0.f : l[g];
+      }
+      part_acc[base * D + tid] = dead ? 0.f : acc[g];
+    }
+  }
+}
+
+// Combine the split partials: out[b,h,:] = sum_i acc_i*exp(m_i-M) / L.
+// Grid: (B, Hq); block: DECODE_BLOCK (>= D).
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_decode_reduce_kernel(bf16 *__restrict__ out,
+                                const float *__restrict__ part_m,
+                                const float *__restrict__ part_l,
+                                const float *__restrict__ part_acc,
+                                int Hq, int D, int NS) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  const int tid = threadIdx.x;
+  const long base = ((long)b * Hq + h) * NS;
+
+  float M = -INFINITY;
+  for (int i = tid; i < NS; i += blockDim.x)
+    M = fmaxf(M, part_m[base + i]);
+  __shared__ float scratch[8];
+  M = block_max(M, scratch);
+
+  float L = 0.f;
+  for (int i = tid; i < NS; i += blockDim.x) {
+    float mi = part_m[base + i];
+    L += (mi == -INFINITY) ? 0.f : part_l[base + i] * __expf(mi - M);
+  }
+  L = block_sum(L, scratch);
+
+  if (tid < D) {
+    float o = 0.f;
+    for (int i = 0; i < NS; ++i) {
+      float mi = part_m[base + i];
+      if (mi == -INFINITY) continue;
+      o = fmaf(part_acc[(base + i) * D + tid], __expf(mi - M), o);
+    }
+    out[((long)b * Hq + h) * D + tid] = f2bf(L > 0.f ? o / L : 0.f);
+  }
+}

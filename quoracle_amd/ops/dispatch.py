@@ -55,10 +55,26 @@ def kv_append(kcache, vcache, k, v, slots) -> None:
 
 
 def paged_attn_decode(out, q, kcache, vcache, block_tables, ctx_lens,
-                      scale: float):
+                      scale: float, max_ctx: int = 0):
+    """GQA decode attention.  On GPU, long contexts take the flash-decoding
+    split path (context partitioned over NS workgroups + combine) so small
+    agent batches still fill the 256 CUs; short contexts use the single-pass
+    kernel (no workspace traffic)."""
     if q.is_cuda:
-        ext().paged_attn_decode(out, q, kcache, vcache, block_tables,
-                                ctx_lens, scale)
+        B, Hq, D = q.shape
+        if max_ctx >= 1024:
+            ns = min(32, max(2, max_ctx // 256))
+            part_m = torch.empty((B, Hq, ns), dtype=torch.float32,
+                                 device=q.device)
+            part_l = torch.empty_like(part_m)
+            part_acc = torch.empty((B, Hq, ns, D), dtype=torch.float32,
+                                   device=q.device)
+            ext().paged_attn_decode_split(out, q, kcache, vcache,
+                                          block_tables, ctx_lens, scale,
+                                          part_m, part_l, part_acc)
+        else:
+            ext().paged_attn_decode(out, q, kcache, vcache, block_tables,
+                                    ctx_lens, scale)
         return out
     return ref.paged_attn_decode(out, q, kcache, vcache, block_tables,
                                  ctx_lens, scale)
