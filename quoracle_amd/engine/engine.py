@@ -33,6 +33,7 @@ import torch
 
 from ..models import LlamaModel, get_config
 from .api import Engine, GenerateRequest, GenerateResult, MIN_OUTPUT_TOKENS
+from .graphs import DecodeGraphs
 from .kv_cache import BlockManager, OutOfBlocks, Session, SessionCache
 from .sampler import ActionGrammar, Sampler, SamplingParams
 from .tokenizer import ByteTokenizer, EOS
@@ -84,6 +85,11 @@ class _HostedModel:
         # overlap on the GPU instead of serializing through the engine loop
         self.stream = (torch.cuda.Stream(device)
                        if device.type == "cuda" else None)
+        # hipGraph-captured decode step (scratch block absorbs pad-row writes)
+        scratch = self.mgr.alloc(1)[0]
+        maxb = min(kv_blocks, (self.cfg.max_context + BLOCK_SIZE - 1)
+                   // BLOCK_SIZE)
+        self.graphs = DecodeGraphs(self.model, self.kv, device, maxb, scratch)
 
     def stream_ctx(self):
         import contextlib
@@ -351,6 +357,20 @@ class LocalEngine(Engine):
             bt_rows.append(list(seq.session.blocks))
             ctx_lens.append(cached + 1)
         n_decode = len(ctx_lens)
+
+        # pure-decode steps replay the hipGraph-captured step when possible
+        if n_decode and not prefill:
+            sample_seqs = [s for s in decode if s not in failed]
+            with hm.stream_ctx():
+                logits = hm.graphs.run(tokens, positions, slots,
+                                       bt_rows, ctx_lens)
+            if logits is not None:
+                st = self.stats
+                st["engine_steps"] += 1
+                st["graph_steps"] = st.get("graph_steps", 0) + 1
+                st["forward_tokens"] += n_decode
+                st["decode_tokens"] += n_decode
+                return (sample_seqs, logits, failed)
 
         tile_q0: List[int] = []
         tile_qn: List[int] = []

@@ -1,0 +1,140 @@
+"""hipGraph capture of the steady-state decode step.
+
+The decode hot loop is hundreds of tiny kernels (32 layers x ~8 launches for
+a 1-token-per-sequence batch): eager launch costs ~15 ms of host time per
+step — far more than the GPU work.  Capture the whole decode forward +
+logits GEMM into a hipGraph per (model, batch-bucket) with static
+input/output buffers; each step then copies the new token/position/slot/
+block-table values into the static buffers and replays one graph
+("hipGraph capture of the steady-state decode step").
+
+Batch sizes are bucketed to powers of two; rows beyond the real batch are
+padded with ctx_len=1 pointing at a reserved scratch block, so padding work
+is negligible and harmless (its KV writes land in the scratch block).
+
+Prefill steps stay eager: they are few, large, and shape-diverse.
+"""
+
+from __future__ import annotations
+
+import sys
+from typing import Dict, List, Optional
+
+import torch
+
+from ..models.llama import ForwardBatch
+
+_BUCKETS = [1, 2, 4, 8, 16, 32, 64]
+NS_GRAPH = 32          # fixed split count inside the graph
+
+
+def bucket_for(n: int) -> Optional[int]:
+    for b in _BUCKETS:
+        if n <= b:
+            return b
+    return None
+
+
+class DecodeGraphs:
+    def __init__(self, model, kv, device: torch.device, max_blocks_per_seq: int,
+                 scratch_block: int):
+        self.model = model
+        self.kv = kv
+        self.device = device
+        self.maxb = max_blocks_per_seq
+        self.scratch_block = scratch_block
+        self.graphs: Dict[int, dict] = {}
+        self.enabled = device.type == "cuda" and not bool(
+            __import__("os").environ.get("QUORACLE_NO_GRAPHS"))
+        self.pool = None
+
+    # -- capture -------------------------------------------------------------
+
+    def _capture(self, bucket: int) -> Optional[dict]:
+        dev = self.device
+        V = self.model.cfg.vocab_size
+        bufs = {
+            "tokens": torch.zeros(bucket, dtype=torch.int32, device=dev),
+            "positions": torch.zeros(bucket, dtype=torch.int32, device=dev),
+            "slots": torch.full((bucket,),
+                                self.scratch_block * self.kv.block_size,
+                                dtype=torch.int32, device=dev),
+            "block_tables": torch.full((bucket, self.maxb), self.scratch_block,
+                                       dtype=torch.int32, device=dev),
+            "ctx_lens": torch.ones(bucket, dtype=torch.int32, device=dev),
+        }
+        rows = torch.arange(bucket, dtype=torch.long, device=dev)
+
+        def _forward():
+            batch = ForwardBatch(
+                tokens=bufs["tokens"], positions=bufs["positions"],
+                slots=bufs["slots"], block_tables=bufs["block_tables"],
+                n_decode=bucket, ctx_lens=bufs["ctx_lens"],
+                max_ctx=1 << 30)     # force the split decode path (NS fixed)
+            hidden = self.model.forward(batch, self.kv)
+            return self.model.compute_logits(hidden, rows)
+
+        try:
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream(dev)
+            s.wait_stream(torch.cuda.current_stream(dev))
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    logits = _forward()
+            torch.cuda.current_stream(dev).wait_stream(s)
+            torch.cuda.synchronize(dev)
+
+            graph = torch.cuda.CUDAGraph()
+            if self.pool is None:
+                with torch.cuda.graph(graph):
+                    logits = _forward()
+                self.pool = graph.pool()
+            else:
+                with torch.cuda.graph(graph, pool=self.pool):
+                    logits = _forward()
+        except Exception as exc:  # noqa: BLE001 — graphs are an optimization
+            print(f"[graphs] capture failed for bucket {bucket}: {exc}; "
+                  f"falling back to eager decode", file=sys.stderr, flush=True)
+            self.enabled = False
+            return None
+        entry = {"graph": graph, "bufs": bufs, "logits": logits}
+        self.graphs[bucket] = entry
+        return entry
+
+    # -- replay --------------------------------------------------------------
+
+    def run(self, tokens: List[int], positions: List[int], slots: List[int],
+            bt_rows: List[List[int]], ctx_lens: List[int]):
+        """Returns logits[:B] or None if graphs are unavailable."""
+        if not self.enabled:
+            return None
+        B = len(tokens)
+        bucket = bucket_for(B)
+        if bucket is None:
+            return None
+        if max(len(r) for r in bt_rows) > self.maxb:
+            return None
+        entry = self.graphs.get(bucket) or self._capture(bucket)
+        if entry is None:
+            return None
+        bufs = entry["bufs"]
+        pad_slot = self.scratch_block * self.kv.block_size
+        t = torch.zeros(bucket, dtype=torch.int32)
+        t[:B] = torch.tensor(tokens, dtype=torch.int32)
+        bufs["tokens"].copy_(t, non_blocking=True)
+        p = torch.zeros(bucket, dtype=torch.int32)
+        p[:B] = torch.tensor(positions, dtype=torch.int32)
+        bufs["positions"].copy_(p, non_blocking=True)
+        sl = torch.full((bucket,), pad_slot, dtype=torch.int32)
+        sl[:B] = torch.tensor(slots, dtype=torch.int32)
+        bufs["slots"].copy_(sl, non_blocking=True)
+        cl = torch.ones(bucket, dtype=torch.int32)
+        cl[:B] = torch.tensor(ctx_lens, dtype=torch.int32)
+        bufs["ctx_lens"].copy_(cl, non_blocking=True)
+        bt = torch.full((bucket, self.maxb), self.scratch_block,
+                        dtype=torch.int32)
+        for i, r in enumerate(bt_rows):
+            bt[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+        bufs["block_tables"].copy_(bt, non_blocking=True)
+        entry["graph"].replay()
+        return entry["logits"][:B]
