@@ -85,12 +85,15 @@ class DecodeGraphs:
             torch.cuda.synchronize(dev)
 
             graph = torch.cuda.CUDAGraph()
+            # thread_local: the orchestrator thread (embedder, monitors) may
+            # issue work on other streams while this capture is open
             if self.pool is None:
-                with torch.cuda.graph(graph):
+                with torch.cuda.graph(graph, capture_error_mode="thread_local"):
                     logits = _forward()
                 self.pool = graph.pool()
             else:
-                with torch.cuda.graph(graph, pool=self.pool):
+                with torch.cuda.graph(graph, pool=self.pool,
+                                      capture_error_mode="thread_local"):
                     logits = _forward()
         except Exception as exc:  # noqa: BLE001 — graphs are an optimization
             print(f"[graphs] capture failed for bucket {bucket}: {exc}; "
