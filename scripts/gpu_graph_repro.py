@@ -1,5 +1,7 @@
 """Isolate the hipGraph decode crash: tiny model, eager vs graph logits."""
-import torch, sys
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
 from quoracle_amd.models import LlamaModel
 from quoracle_amd.models.llama import ForwardBatch
 from quoracle_amd.engine.graphs import DecodeGraphs
