@@ -64,6 +64,7 @@ class DecodeGraphs:
             "ctx_lens": torch.ones(bucket, dtype=torch.int32, device=dev),
         }
         rows = torch.arange(bucket, dtype=torch.long, device=dev)
+        scope = __import__("os").environ.get("QUORACLE_GRAPH_SCOPE", "full")
 
         def _forward():
             batch = ForwardBatch(
@@ -72,6 +73,8 @@ class DecodeGraphs:
                 n_decode=bucket, ctx_lens=bufs["ctx_lens"],
                 max_ctx=1 << 30)     # force the split decode path (NS fixed)
             hidden = self.model.forward(batch, self.kv)
+            if scope == "noheads":
+                return hidden
             return self.model.compute_logits(hidden, rows)
 
         try:
@@ -100,7 +103,8 @@ class DecodeGraphs:
                   f"falling back to eager decode", file=sys.stderr, flush=True)
             self.enabled = False
             return None
-        entry = {"graph": graph, "bufs": bufs, "logits": logits}
+        entry = {"graph": graph, "bufs": bufs, "logits": logits,
+                 "rows": rows, "scope": scope}
         self.graphs[bucket] = entry
         return entry
 
@@ -140,4 +144,7 @@ class DecodeGraphs:
             bt[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
         bufs["block_tables"].copy_(bt, non_blocking=True)
         entry["graph"].replay()
+        if entry["scope"] == "noheads":
+            return self.model.compute_logits(entry["logits"],
+                                             entry["rows"])[:B]
         return entry["logits"][:B]

@@ -2,6 +2,8 @@
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
+if os.environ.get("QUORACLE_BLAS") == "rocblas":
+    torch.backends.cuda.preferred_blas_library("cublas")
 from quoracle_amd.models import LlamaModel
 from quoracle_amd.models.llama import ForwardBatch
 from quoracle_amd.engine.graphs import DecodeGraphs
@@ -40,6 +42,7 @@ if stage == "single":
 
     pos = CTX
     for i in range(40):
+        print(f"replay {i}", flush=True) if i < 5 else None
         tok, slot, ctx = [int(7 + i)], [blocks[pos // 16] * 16 + pos % 16], [pos + 1]
         logits = graphs.run(tok, [pos], slot, [blocks], ctx)
         assert logits is not None, "graphs disabled"
