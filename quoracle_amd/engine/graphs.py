@@ -47,6 +47,9 @@ class DecodeGraphs:
         self.enabled = device.type == "cuda" and not bool(
             __import__("os").environ.get("QUORACLE_NO_GRAPHS"))
         self.pool = None
+        # capture only happens while the GPU is quiescent (engine start);
+        # a missing bucket at run time falls back to eager, never captures
+        self.allow_capture = False
 
     # -- capture -------------------------------------------------------------
 
@@ -108,6 +111,19 @@ class DecodeGraphs:
         self.graphs[bucket] = entry
         return entry
 
+    def precapture(self, buckets=(1, 2, 4, 8, 16)) -> None:
+        """Capture the decode graphs for the given batch buckets while the
+        device is quiescent (called from engine start, before any traffic)."""
+        if not self.enabled:
+            return
+        self.allow_capture = True
+        try:
+            for b in buckets:
+                if b not in self.graphs and self._capture(b) is None:
+                    break
+        finally:
+            self.allow_capture = False
+
     # -- replay --------------------------------------------------------------
 
     def run(self, tokens: List[int], positions: List[int], slots: List[int],
@@ -121,9 +137,13 @@ class DecodeGraphs:
             return None
         if max(len(r) for r in bt_rows) > self.maxb:
             return None
-        entry = self.graphs.get(bucket) or self._capture(bucket)
+        entry = self.graphs.get(bucket)
         if entry is None:
-            return None
+            if not self.allow_capture:
+                return None
+            entry = self._capture(bucket)
+            if entry is None:
+                return None
         bufs = entry["bufs"]
         pad_slot = self.scratch_block * self.kv.block_size
         t = torch.zeros(bucket, dtype=torch.int32)

@@ -86,11 +86,16 @@ elif stage == "engine":
     for r in reqs:
         eng._submit(r, ("cb", lambda res: (done.append(res),
                                            len(done) == 6 and ev.set())))
+    eng.precapture_graphs((1, 2))
     import time
-    t0 = time.time()
+    t0 = time.time(); last = t0
     while not ev.is_set():
         eng.step()
-        if time.time() - t0 > 300:
+        now = time.time()
+        if now - last > 10:
+            last = now
+            print(f"hb t={now - t0:.0f}s stats={eng.stats}", flush=True)
+        if now - t0 > 280:
             print("TIMEOUT", flush=True); sys.exit(2)
     for r in done:
         print(r.model_key, r.error, r.output_tokens, flush=True)
