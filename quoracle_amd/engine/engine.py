@@ -373,19 +373,28 @@ class LocalEngine(Engine):
             ctx_lens.append(cached + 1)
         n_decode = len(ctx_lens)
 
-        # pure-decode steps replay the hipGraph-captured step when possible
+        # pure-decode steps replay the hipGraph-captured step when possible.
+        # Replay + its input copies run on the DEFAULT stream: hipGraph
+        # replay launched from a side stream deadlocks on ROCm 7.2 (observed
+        # on MI355X; graph kernels never complete), so graph steps serialize
+        # across models while eager prefill keeps per-model streams.
         if n_decode and not prefill:
             sample_seqs = [s for s in decode if s not in failed]
-            with hm.stream_ctx():
-                logits = hm.graphs.run(tokens, positions, slots,
-                                       bt_rows, ctx_lens)
+            if hm.stream is not None:
+                # order after this model's eager (prefill) stream, and make
+                # that stream wait for the replay before later prefills
+                torch.cuda.current_stream(self.device).wait_stream(hm.stream)
+            logits = hm.graphs.run(tokens, positions, slots,
+                                   bt_rows, ctx_lens)
+            if logits is not None and hm.stream is not None:
+                hm.stream.wait_stream(torch.cuda.current_stream(self.device))
             if logits is not None:
                 st = self.stats
                 st["engine_steps"] += 1
                 st["graph_steps"] = st.get("graph_steps", 0) + 1
                 st["forward_tokens"] += n_decode
                 st["decode_tokens"] += n_decode
-                return (sample_seqs, logits, failed)
+                return (sample_seqs, logits, failed, True)
 
         tile_q0: List[int] = []
         tile_qn: List[int] = []
@@ -456,13 +465,15 @@ class LocalEngine(Engine):
                   f"T={len(tokens)} dec={n_decode} active={len(hm.active)} "
                   f"launch_ms={(time.monotonic() - t_fwd) * 1e3:.1f}",
                   file=sys.stderr, flush=True)
-        return (sample_seqs, logits, failed)
+        return (sample_seqs, logits, failed, False)
 
     def _sample_model(self, hm: _HostedModel, sample_seqs, logits,
-                      failed) -> None:
+                      failed, on_default_stream=False) -> None:
+        import contextlib
         done: List[_Seq] = list(failed)
         if logits is not None:
-            with hm.stream_ctx():
+            with (contextlib.nullcontext() if on_default_stream
+                  else hm.stream_ctx()):
                 ids = hm.sampler.sample(
                     logits, [s.params for s in sample_seqs],
                     [s.grammar for s in sample_seqs],
