@@ -178,15 +178,16 @@ class Sampler:
         Returns emitted token ids (grammar-adjusted)."""
         R = logits.shape[0]
         safe = _MaskCache.get(self.vocab_size, logits.device)
-        # Build the masked/temperature-scaled distribution per row, then one
-        # multinomial per row (R is small: decode batch of agents).
-        emitted: List[int] = []
+        # Launch every row's sampling asynchronously (forced-token rows need
+        # no GPU work at all), then ONE host sync for the whole batch.
+        emitted: List[Optional[int]] = [None] * R
+        pending: List[Tuple[int, torch.Tensor]] = []
         for r in range(R):
             p, g = params[r], grammars[r]
             if g is not None:
                 op = g.current()
                 if op[0] == FORCED:
-                    emitted.append(g.advance(0))
+                    emitted[r] = g.advance(0)
                     continue
                 row = logits[r] + safe        # FREE and CHOICE: printable only
             else:
@@ -195,8 +196,13 @@ class Sampler:
             probs = torch.softmax(row / t, dim=-1)
             if p.top_p < 1.0:
                 probs = _top_p_filter(probs, p.top_p)
-            idx = int(torch.multinomial(probs, 1, generator=generators[r]))
-            emitted.append(g.advance(idx) if g is not None else idx)
+            pending.append((r, torch.multinomial(probs, 1,
+                                                 generator=generators[r])))
+        if pending:
+            ids = torch.cat([t for _, t in pending]).cpu()   # single sync
+            for (r, _), idx in zip(pending, ids.tolist()):
+                g = grammars[r]
+                emitted[r] = g.advance(idx) if g is not None else idx
         return emitted
 
 
