@@ -432,6 +432,7 @@ class LocalEngine(Engine):
         sample_seqs: List[_Seq] = [s for s in decode if s not in failed]
 
         row = n_decode
+        max_kv = 0
         for seq, n in prefill:
             cached = len(seq.session.token_ids)
             chunk = seq.known[cached:cached + n]
@@ -451,6 +452,7 @@ class LocalEngine(Engine):
             if cached + n == len(seq.known):
                 sample_rows.append(row + n - 1)
                 sample_seqs.append(seq)
+            max_kv = max(max_kv, cached + n)
             row += n
 
         if not tokens:
@@ -475,6 +477,7 @@ class LocalEngine(Engine):
                 n_decode=n_decode,
                 ctx_lens=t32(ctx_lens) if n_decode else None,
                 max_ctx=max(ctx_lens) if ctx_lens else 0,
+                max_kv=max_kv,
                 tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
                 tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
             hidden = hm.model.forward(batch, hm.kv)

@@ -45,6 +45,7 @@ class ForwardBatch:
     n_decode: int = 0
     ctx_lens: Optional[torch.Tensor] = None   # [n_decode] int32 (incl. new tok)
     max_ctx: int = 0                          # host-known max(ctx_lens)
+    max_kv: int = 0                           # host-known max tile kv extent
     # prefill tiles (QT=16 query rows each); empty tensors when no prefill
     tile_q0: Optional[torch.Tensor] = None    # [ntiles] int32 row into tokens
     tile_qn: Optional[torch.Tensor] = None
@@ -171,7 +172,7 @@ class LlamaModel:
                 D.paged_attn_prefill(
                     attn_out, q, kv.k[li], kv.v[li], batch.block_tables,
                     batch.tile_q0, batch.tile_qn, batch.tile_seq,
-                    batch.tile_pos0, self.scale)
+                    batch.tile_pos0, self.scale, max_kv=batch.max_kv)
             proj = attn_out.view(T, cfg.q_dim) @ layer["wo"]
             D.rmsnorm(h, proj, res, layer["ffn_norm"], cfg.rmsnorm_eps)
             if cfg.is_moe:

@@ -446,3 +446,184 @@ paged_attn_decode_reduce_kernel(bf16 *__restrict__ out,
     out[((long)b * Hq + h) * D + tid] = f2bf(L > 0.f ? o / L : 0.f);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Split prefill: small late-conversation chunks (grammar forced-runs, few
+// tiles) leave the (ntiles, Hq) grid nearly empty while every tile walks the
+// WHOLE cached context serially.  Partition the KV walk across NS workgroups
+// per (tile, head) exactly like the decode split path.
+//   part_m/part_l: [ntiles, Hq, NS, QT] f32
+//   part_acc:      [ntiles, Hq, NS, QT, D] f32
+// Grid: (ntiles, Hq, NS); block: DECODE_BLOCK.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_prefill_split_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int D,
+    int BS, int MAXB, int GQ, int NS) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int split = blockIdx.z;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+  const int chunks = (kv_limit + CHUNK - 1) / CHUNK;
+  const int per = (chunks + NS - 1) / NS;
+  const int t0 = split * per * CHUNK;
+  const int t1 = min(kv_limit, (split + 1) * per * CHUNK);
+
+  __shared__ float q_s[QT * 128];
+  __shared__ float p_s[QT][CHUNK];
+  __shared__ float scratch[8];
+
+  for (int i = tid; i < qn * D; i += blockDim.x) {
+    int qi = i / D, d = i % D;
+    q_s[qi * D + d] = bf2f(q[((long)(q0 + qi) * Hq + h) * D + d]) * scale;
+  }
+  __syncthreads();
+
+  float m[QT], l[QT], acc[QT];
+#pragma unroll
+  for (int qi = 0; qi < QT; ++qi) {
+    m[qi] = -INFINITY;
+    l[qi] = 0.f;
+    acc[qi] = 0.f;
+  }
+
+  const long panel_stride = (long)Hkv * BS * D;
+  for (int start = t0; start < t1; start += CHUNK) {
+    const int clen = min(CHUNK, t1 - start);
+    if (tid < clen) {
+      const int token = start + tid;
+      const long blk = bt[(long)seq * MAXB + token / BS];
+      const bf16 *krow =
+          kc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+      float dots[QT];
+#pragma unroll
+      for (int qi = 0; qi < QT; ++qi) dots[qi] = 0.f;
+      for (int d8 = 0; d8 < D / 8; ++d8) {
+        uint4 kv = reinterpret_cast<const uint4 *>(krow)[d8];
+        float kf[8];
+        unpack_bf16x2(kv.x, kf[0], kf[1]);
+        unpack_bf16x2(kv.y, kf[2], kf[3]);
+        unpack_bf16x2(kv.z, kf[4], kf[5]);
+        unpack_bf16x2(kv.w, kf[6], kf[7]);
+#pragma unroll
+        for (int qi = 0; qi < QT; ++qi) {
+          if (qi >= qn) break;
+          const float *qg = q_s + qi * D + d8 * 8;
+#pragma unroll
+          for (int k = 0; k < 8; ++k)
+            dots[qi] = fmaf(kf[k], qg[k], dots[qi]);
+        }
+      }
+#pragma unroll
+      for (int qi = 0; qi < QT; ++qi) {
+        if (qi >= qn) break;
+        p_s[qi][tid] = (token <= pos0 + qi) ? dots[qi] : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int qi = 0; qi < QT; ++qi) {
+      if (qi >= qn) break;
+      float mine = (tid < clen) ? p_s[qi][tid] : -INFINITY;
+      float cmax = block_max(mine, scratch);
+      if (cmax == -INFINITY) {
+        if (tid < clen) p_s[qi][tid] = 0.f;
+        __syncthreads();
+        continue;
+      }
+      float mn = fmaxf(m[qi], cmax);
+      float alpha = (m[qi] == -INFINITY) ? 0.f : __expf(m[qi] - mn);
+      float p = (tid < clen && p_s[qi][tid] != -INFINITY)
+                    ? __expf(p_s[qi][tid] - mn) : 0.f;
+      if (tid < clen) p_s[qi][tid] = p;
+      float psum = block_sum(p, scratch);
+      l[qi] = l[qi] * alpha + psum;
+      acc[qi] *= alpha;
+      m[qi] = mn;
+    }
+    __syncthreads();
+
+    if (tid < D) {
+      for (int i = 0; i < clen; ++i) {
+        const int token = start + i;
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const bf16 *vrow =
+            vc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
+        const float v = bf2f(vrow[tid]);
+#pragma unroll
+        for (int qi = 0; qi < QT; ++qi) {
+          if (qi >= qn) break;
+          acc[qi] = fmaf(p_s[qi][i], v, acc[qi]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const bool dead = (t0 >= kv_limit);
+  if (tid < D) {
+#pragma unroll
+    for (int qi = 0; qi < QT; ++qi) {
+      if (qi >= qn) break;
+      const long base = (((long)tile * Hq + h) * NS + split) * QT + qi;
+      if (tid == 0) {
+        part_m[base] = dead ? -INFINITY : m[qi];
+        part_l[base] = dead ? 0.f : l[qi];
+      }
+      part_acc[base * D + tid] = dead ? 0.f : acc[qi];
+    }
+  }
+}
+
+// Combine prefill split partials.  Grid: (ntiles, Hq); block: DECODE_BLOCK.
+extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
+paged_attn_prefill_reduce_kernel(bf16 *__restrict__ out,
+                                 const float *__restrict__ part_m,
+                                 const float *__restrict__ part_l,
+                                 const float *__restrict__ part_acc,
+                                 const int *__restrict__ tile_q0,
+                                 const int *__restrict__ tile_qn,
+                                 int Hq, int D, int NS) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  __shared__ float scratch[8];
+
+  for (int qi = 0; qi < qn; ++qi) {
+    const long base0 = (((long)tile * Hq + h) * NS) * QT + qi;
+    float M = -INFINITY;
+    for (int i = tid; i < NS; i += blockDim.x)
+      M = fmaxf(M, part_m[base0 + (long)i * QT]);
+    M = block_max(M, scratch);
+    float L = 0.f;
+    for (int i = tid; i < NS; i += blockDim.x) {
+      float mi = part_m[base0 + (long)i * QT];
+      L += (mi == -INFINITY) ? 0.f : part_l[base0 + (long)i * QT] * __expf(mi - M);
+    }
+    L = block_sum(L, scratch);
+    if (tid < D) {
+      float o = 0.f;
+      for (int i = 0; i < NS; ++i) {
+        float mi = part_m[base0 + (long)i * QT];
+        if (mi == -INFINITY) continue;
+        o = fmaf(part_acc[(base0 + (long)i * QT) * D + tid], __expf(mi - M), o);
+      }
+      out[((long)(q0 + qi) * Hq + h) * D + tid] = f2bf(L > 0.f ? o / L : 0.f);
+    }
+    __syncthreads();
+  }
+}

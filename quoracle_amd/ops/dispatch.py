@@ -81,8 +81,29 @@ def paged_attn_decode(out, q, kcache, vcache, block_tables, ctx_lens,
 
 
 def paged_attn_prefill(out, q, kcache, vcache, block_tables, tile_q0,
-                       tile_qn, tile_seq, tile_pos0, scale: float):
+                       tile_qn, tile_seq, tile_pos0, scale: float,
+                       max_kv: int = 0):
+    """Chunked causal prefill over the paged cache.  Few tiles over a long
+    cached context (grammar forced-run chunks, incremental history prefill)
+    take the context-split path so the walk parallelizes across the chip."""
     if q.is_cuda:
+        ntiles = tile_q0.shape[0]
+        Hq = q.shape[1]
+        D = q.shape[2]
+        # target ~2048 workgroups; split when the plain grid is too empty
+        ns = max(1, min(16, 2048 // max(1, ntiles * Hq)))
+        if ns > 1 and max_kv >= 1024:
+            QT = 16
+            part_m = torch.empty((ntiles, Hq, ns, QT), dtype=torch.float32,
+                                 device=q.device)
+            part_l = torch.empty_like(part_m)
+            part_acc = torch.empty((ntiles, Hq, ns, QT, D),
+                                   dtype=torch.float32, device=q.device)
+            ext().paged_attn_prefill_split(out, q, kcache, vcache,
+                                           block_tables, tile_q0, tile_qn,
+                                           tile_seq, tile_pos0, scale,
+                                           part_m, part_l, part_acc)
+            return out
         ext().paged_attn_prefill(out, q, kcache, vcache, block_tables,
                                  tile_q0, tile_qn, tile_seq, tile_pos0, scale)
         return out
