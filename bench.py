@@ -30,7 +30,7 @@ import time
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=4)
+    p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--pool-size", type=int, default=3)
