@@ -1,0 +1,206 @@
+// MFMA-tiled paged prefill attention for gfx950 (CDNA4).
+//
+// The VALU prefill kernel profiles at 71% of bench GPU time
+// (profiles/r01_bench_kernel_stats.csv): dot products on the vector ALU at
+// ~7% efficiency.  This kernel moves QK^T and P·V onto the matrix cores
+// with the guide's attention structure (cdna_hip_programming.md §3, §5.5):
+//
+//   * one workgroup (4 waves, 256 threads) per (16-query tile, q-head),
+//     walking the KV context in 64-key chunks;
+//   * K staged row-major and V staged TRANSPOSED in LDS with +8 element row
+//     padding so every MFMA fragment read is bank-conflict-free;
+//   * S = Q·K^T via v_mfma_f32_16x16x32_bf16 (each wave owns one 16-key
+//     block, K-loop over D=128 in 4 steps);
+//   * online softmax in LDS (persistent m/l per query row), P written back
+//     as bf16 and re-read in the A-fragment layout;
+//   * O += P·V accumulated in C fragments across chunks with per-chunk
+//     alpha rescaling (textbook order: P of a chunk is exponentiated only
+//     after the max decision that covers it — T13 hazard avoided).
+//
+// Fragment maps (cdna4_isa.md §10, 16x16x32 bf16):
+//   A: row = lane&15,           k = (lane>>4)*8 + i   (8 contiguous)
+//   B: col = lane&15,           k = (lane>>4)*8 + i
+//   C/D: col = lane&15,         row = (lane>>4)*4 + reg
+//
+// Requires D == 128; host dispatch falls back to the VALU kernels
+// otherwise (quoracle_amd/ops/dispatch.py).
+
+#include "common.h"
+
+#define MF_KCHUNK 64
+#define MF_QT 16
+#define MF_D 128
+// LDS row pads keep fragment reads conflict-free (see dispatch notes)
+#define KP (MF_D + 8)        // k_s / q_s row stride (elements)
+#define VP (MF_KCHUNK + 8)   // vt_s / p_s row stride
+#define SP (MF_KCHUNK + 4)   // s_s row stride (f32 words)
+
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+extern "C" __global__ void __launch_bounds__(256)
+paged_attn_prefill_mfma_kernel(
+    bf16 *__restrict__ out, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+
+  __shared__ bf16 q_s[MF_QT * KP];
+  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ float s_s[MF_QT * SP];
+  __shared__ bf16 p_s[MF_QT * VP];
+  __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
+
+  // ---- load the Q tile (rows >= qn zeroed) --------------------------------
+  for (int i = tid; i < MF_QT * MF_D / 8; i += 256) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
+  if (tid < MF_QT) {
+    m_s[tid] = -INFINITY;
+    l_s[tid] = 0.f;
+  }
+  __syncthreads();
+
+  // O accumulator fragments: wave w owns output cols [w*32, w*32+32)
+  f32x4_t o_acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t o_acc1 = {0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int a_row = lane & 15;           // A/B col or row
+  const int a_koff = (lane >> 4) * 8;    // k offset within 32-wide K slice
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+
+    // ---- stage K (row-major) and V (transposed) ---------------------------
+    for (int i = tid; i < MF_KCHUNK * MF_D / 8; i += 256) {
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kval = make_uint4(0, 0, 0, 0);
+      uint4 vval = make_uint4(0, 0, 0, 0);
+      if (key < clen) {
+        const int token = start + key;
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off = blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kval = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vval = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kval;
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vval);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+    }
+    __syncthreads();
+
+    // ---- S = Q·K^T (wave w: key block w*16..w*16+15) ----------------------
+    {
+      f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            q_s + a_row * KP + kk * 32 + a_koff);
+        bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = c_row0 + r;
+        const int col = wave * 16 + c_col;
+        const int token = start + col;
+        const bool ok = (token <= pos0 + min(row, qn - 1)) && (col < clen);
+        s_s[row * SP + col] = ok ? s_acc[r] * scale : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+    // ---- online softmax (16 threads per query row) ------------------------
+    {
+      const int row = tid >> 4;
+      const int sub = tid & 15;
+      float v0 = s_s[row * SP + sub];
+      float v1 = s_s[row * SP + sub + 16];
+      float v2 = s_s[row * SP + sub + 32];
+      float v3 = s_s[row * SP + sub + 48];
+      float mymax = fmaxf(fmaxf(v0, v1), fmaxf(v2, v3));
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        mymax = fmaxf(mymax, __shfl_xor(mymax, w, 16));
+      const float m_old = m_s[row];
+      const float mn = fmaxf(m_old, mymax);
+      const float alpha = (m_old == -INFINITY) ? 0.f : __expf(m_old - mn);
+      float p0 = (v0 == -INFINITY) ? 0.f : __expf(v0 - mn);
+      float p1 = (v1 == -INFINITY) ? 0.f : __expf(v1 - mn);
+      float p2 = (v2 == -INFINITY) ? 0.f : __expf(v2 - mn);
+      float p3 = (v3 == -INFINITY) ? 0.f : __expf(v3 - mn);
+      float psum = p0 + p1 + p2 + p3;
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        psum += __shfl_xor(psum, w, 16);
+      if (sub == 0) {
+        l_s[row] = l_s[row] * alpha + psum;
+        m_s[row] = mn;
+        alpha_s[row] = alpha;
+      }
+      p_s[row * VP + sub] = f2bf(p0);
+      p_s[row * VP + sub + 16] = f2bf(p1);
+      p_s[row * VP + sub + 32] = f2bf(p2);
+      p_s[row * VP + sub + 48] = f2bf(p3);
+    }
+    __syncthreads();
+
+    // ---- O rescale + O += P·V --------------------------------------------
+    {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float a = alpha_s[c_row0 + r];
+        o_acc0[r] *= a;
+        o_acc1[r] *= a;
+      }
+#pragma unroll
+      for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            p_s + a_row * VP + kk * 32 + a_koff);
+        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
+        o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
+      }
+    }
+    __syncthreads();   // next chunk restages k_s/vt_s
+  }
+
+  // ---- epilogue -----------------------------------------------------------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = c_row0 + r;
+    if (row >= qn) continue;
+    const float denom = l_s[row] > 0.f ? l_s[row] : 1.f;
+    out[((long)(q0 + row) * Hq + h) * MF_D + wave * 32 + c_col] =
+        f2bf(o_acc0[r] / denom);
+    out[((long)(q0 + row) * Hq + h) * MF_D + wave * 32 + 16 + c_col] =
+        f2bf(o_acc1[r] / denom);
+  }
+}

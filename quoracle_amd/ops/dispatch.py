@@ -10,6 +10,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 
 from . import ext
@@ -90,8 +92,13 @@ def paged_attn_prefill(out, q, kcache, vcache, block_tables, tile_q0,
         ntiles = tile_q0.shape[0]
         Hq = q.shape[1]
         D = q.shape[2]
-        # target ~2048 workgroups; split when the plain grid is too empty
+        # big prefills with D=128 go to the MFMA-tiled kernel (matrix cores)
         ns = max(1, min(16, 2048 // max(1, ntiles * Hq)))
+        if D == 128 and ns == 1 and not os.environ.get("QUORACLE_NO_MFMA_ATTN"):
+            ext().paged_attn_prefill_mfma(out, q, kcache, vcache,
+                                          block_tables, tile_q0, tile_qn,
+                                          tile_seq, tile_pos0, scale)
+            return out
         if ns > 1 and max_kv >= 1024:
             QT = 16
             part_m = torch.empty((ntiles, Hq, ns, QT), dtype=torch.float32,

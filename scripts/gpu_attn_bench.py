@@ -1,0 +1,41 @@
+"""Micro-bench: MFMA vs VALU prefill attention at bench-like shapes."""
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from quoracle_amd import ops
+
+dev = torch.device("cuda:0")
+ops.ext()
+Hq, Hkv, D, BS = 32, 8, 128, 16
+scale = D ** -0.5
+cached, new = 5000, 2048
+total = cached + new
+nb = (total + BS - 1) // BS
+kcache = torch.randn(nb + 1, Hkv, BS, D, device=dev, dtype=torch.bfloat16)
+vcache = torch.randn_like(kcache)
+tables = torch.arange(1, nb + 1, dtype=torch.int32, device=dev).unsqueeze(0)
+q = torch.randn(new, Hq, D, device=dev, dtype=torch.bfloat16)
+out = torch.empty_like(q)
+ntiles = (new + 15) // 16
+t0 = torch.arange(ntiles, dtype=torch.int32, device=dev) * 16
+qn = torch.clamp(torch.full_like(t0, new) - t0, max=16)
+tseq = torch.zeros_like(t0)
+tpos = t0 + cached
+
+def timeit(fn, n=20):
+    fn(); torch.cuda.synchronize()
+    t = time.perf_counter()
+    for _ in range(n): fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t) / n * 1e3
+
+ms_valu = timeit(lambda: ops.ext().paged_attn_prefill(
+    out, q, kcache, vcache, tables, t0, qn, tseq, tpos, scale))
+out_valu = out.clone()
+ms_mfma = timeit(lambda: ops.ext().paged_attn_prefill_mfma(
+    out, q, kcache, vcache, tables, t0, qn, tseq, tpos, scale))
+rel = (out.float() - out_valu.float()).norm() / out_valu.float().norm()
+flops = 2 * 2 * new * (cached + new / 2) * D * Hq
+print(f"VALU prefill: {ms_valu:.3f} ms ({flops/ms_valu/1e9:.1f} TFLOP/s)")
+print(f"MFMA prefill: {ms_mfma:.3f} ms ({flops/ms_mfma/1e9:.1f} TFLOP/s)  "
+      f"speedup {ms_valu/ms_mfma:.1f}x  rel-vs-valu {rel:.4f}")
