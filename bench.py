@@ -37,6 +37,11 @@ def parse_args():
     p.add_argument("--agents-per-gpu", type=int, default=8)
     p.add_argument("--kv-gb", type=float, default=4.0)
     p.add_argument("--device", default=None, help="override (e.g. cpu)")
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree: the whole job is ONE "
+                        "lockstep TP group hosting a '+'-joined model pool "
+                        "(BASELINE config 5: --tp 4 --model "
+                        "llama3-70b+mixtral-8x7b)")
     return p.parse_args()
 
 
@@ -71,6 +76,27 @@ def main():
 
     from quoracle_amd.engine.engine import LocalEngine
 
+    if args.tp > 1:
+        if world != args.tp:
+            raise SystemExit(f"--tp {args.tp} needs WORLD_SIZE={args.tp}")
+        from quoracle_amd.engine.tp_engine import TPEngine, serve_tp_replica
+        from quoracle_amd.parallel.tp import TPContext
+        tp = TPContext(rank, world)
+        keys = [f"{m}#tp" for m in args.model.split("+")]
+        engine = TPEngine(keys, tp, device=device,
+                          kv_gb_per_model=args.kv_gb,
+                          embed_model_key="embed-small" if rank == 0 else None)
+        print(f"[bench t={time.perf_counter() - T_START:.1f}s] rank {rank}: "
+              f"TP{world} engine up, pool {keys}", file=sys.stderr, flush=True)
+        if rank != 0:
+            serve_tp_replica(engine)
+            return
+        result = asyncio.run(orchestrate(args, engine, device, 1,
+                                         pool_keys=keys,
+                                         parallelism=f"tp{world}"))
+        print(json.dumps(result), flush=True)
+        return
+
     local_keys = rank_model_keys(args.model, args.pool_size, rank)
     engine = LocalEngine(
         local_keys, device=device, kv_gb_per_model=args.kv_gb,
@@ -88,7 +114,8 @@ def main():
     print(json.dumps(result), flush=True)
 
 
-async def orchestrate(args, engine, device, world):
+async def orchestrate(args, engine, device, world, pool_keys=None,
+                      parallelism=None):
     import torch
     from quoracle_amd.agent.core import AgentActor, MESSAGE_TYPES
     from quoracle_amd.agent.state import AgentState
@@ -117,8 +144,12 @@ async def orchestrate(args, engine, device, world):
     for r in range(world):
         runtime.profiles.put(Profile(
             name=f"bench-r{r}", description="bench pool shard",
-            model_pool=rank_model_keys(args.model, args.pool_size, r),
+            model_pool=pool_keys or rank_model_keys(args.model,
+                                                    args.pool_size, r),
             capability_groups=[], max_refinement_rounds=2))
+    if pool_keys:
+        for key in pool_keys:
+            pool.assign(key, engine)
 
     # Build the agent fleet: agents_per_gpu per rank-shard, lockstep-driven
     # (the actor loop is not started — the bench owns cycle timing).
@@ -216,7 +247,7 @@ async def orchestrate(args, engine, device, world):
             "pool_size": args.pool_size,
             "agents_per_gpu": args.agents_per_gpu,
             "agents_total": n_agents,
-            "parallelism": f"pool-sharded dp{world}",
+            "parallelism": parallelism or f"pool-sharded dp{world}",
             "p50_step_latency_ms": round(
                 statistics.median(step_latencies), 2) if step_latencies else None,
             "decisions_completed": decisions,

@@ -72,10 +72,11 @@ class _Seq:
 
 
 class _HostedModel:
-    def __init__(self, key: str, device: torch.device, kv_blocks: int):
+    def __init__(self, key: str, device: torch.device, kv_blocks: int,
+                 tp=None):
         self.key = key
         self.cfg = get_config(key)
-        self.model = LlamaModel(key, device)
+        self.model = LlamaModel(key, device, tp=tp)
         self.kv: KVCache = self.model.new_kv_cache(kv_blocks, BLOCK_SIZE)
         self.mgr = BlockManager(kv_blocks, BLOCK_SIZE)
         self.sessions = SessionCache(self.mgr)
@@ -90,6 +91,10 @@ class _HostedModel:
         maxb = min(kv_blocks, (self.cfg.max_context + BLOCK_SIZE - 1)
                    // BLOCK_SIZE)
         self.graphs = DecodeGraphs(self.model, self.kv, device, maxb, scratch)
+        if tp is not None and tp.world > 1:
+            # RCCL all-reduce under hipGraph capture is unvalidated on this
+            # stack; TP decode stays eager until proven
+            self.graphs.enabled = False
 
     def stream_ctx(self):
         import contextlib
@@ -103,7 +108,8 @@ class LocalEngine(Engine):
                  kv_gb_per_model: float = 4.0,
                  embed_model_key: Optional[str] = "embed-small",
                  kv_blocks_override: Optional[int] = None,
-                 prefill_chunk: int = PREFILL_CHUNK):
+                 prefill_chunk: int = PREFILL_CHUNK,
+                 tp=None):
         self.device = device or torch.device(
             "cuda:0" if torch.cuda.is_available() else "cpu")
         self.tokenizer = ByteTokenizer()
@@ -116,7 +122,7 @@ class LocalEngine(Engine):
             else:
                 per_block = cfg.kv_bytes_per_token() * BLOCK_SIZE
                 blocks = max(8, int(kv_gb_per_model * (1 << 30) / per_block))
-            self.models[key] = _HostedModel(key, self.device, blocks)
+            self.models[key] = _HostedModel(key, self.device, blocks, tp=tp)
         self.embed_model: Optional[LlamaModel] = None
         if embed_model_key:
             self.embed_model = LlamaModel(embed_model_key, self.device)

@@ -1,0 +1,124 @@
+"""Lockstep tensor-parallel engine: one LocalEngine replica per TP rank.
+
+All logits-affecting state is replicated under TP (embeddings, norms,
+residuals and logits are identical on every rank after the per-layer
+all-reduces, and sampling uses per-request seeded generators), so the whole
+continuous-batching engine runs as DETERMINISTIC LOCKSTEP REPLICAS: the
+only synchronization is an admit-broadcast at the top of every engine step
+— no per-token traffic.  The leader rank owns the real request queue and
+result futures; replicas execute identical forwards so the per-layer RCCL
+all-reduces line up (BASELINE config 5).
+
+Exactness requirements (hold by construction):
+  * identical model shards from the same seed (models/llama.py TP sharding),
+  * identical KV block allocation (deterministic free-list),
+  * identical batch composition (this admit-broadcast),
+  * bitwise-replicated logits (all-reduce output identical on all ranks) +
+    per-request seeded generators => identical sampled tokens everywhere.
+"""
+
+from __future__ import annotations
+
+import queue as _q
+import sys
+import time
+from typing import List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+from ..parallel.tp import TPContext
+from .engine import LocalEngine
+
+
+class _StopSentinel:
+    pass
+
+
+class TPEngine(LocalEngine):
+    def __init__(self, model_keys: Sequence[str], tp: TPContext,
+                 device: Optional[torch.device] = None,
+                 control_group=None, **kw):
+        self.tp = tp
+        self.control_group = control_group if control_group is not None \
+            else tp.group
+        self.is_leader = tp.rank == 0
+        self.stop_seen = False
+        self._idle_steps = 0
+        kw.setdefault("tp", tp)
+        super().__init__(model_keys, device=device, **kw)
+
+    def request_stop(self) -> None:
+        """Leader: broadcast shutdown to replicas on the next step."""
+        self._inbox.put(_StopSentinel())
+
+    def step(self) -> bool:
+        if self.tp.world > 1:
+            if self.is_leader:
+                drained = self._drain_inbox()
+                stop = any(isinstance(s, _StopSentinel) for s in drained)
+                seqs = [s for s in drained if not isinstance(s, _StopSentinel)]
+                box = [([s.request for s in seqs], stop)]
+                dist.broadcast_object_list(box, src=0,
+                                           group=self.control_group)
+                for s in seqs:
+                    self._inbox.put(s)
+                if stop:
+                    self.stop_seen = True
+            else:
+                box = [None]
+                dist.broadcast_object_list(box, src=0,
+                                           group=self.control_group)
+                reqs, stop = box[0]
+                if stop:
+                    self.stop_seen = True
+                for req in reqs:
+                    err = self._submit(req, None)
+                    if err is not None:
+                        print(f"[tp-engine r{self.tp.rank}] divergent "
+                              f"submit: {err.error}", file=sys.stderr,
+                              flush=True)
+        worked = super().step()
+        if self.tp.world > 1:
+            # deterministic shared idle backoff: every rank sees the same
+            # (worked, active) state, so all sleep together
+            self._idle_steps = 0 if worked else self._idle_steps + 1
+            if not worked and self._idle_steps > 3:
+                time.sleep(0.004)
+        return worked
+
+    def _drain_inbox(self) -> List:
+        drained = []
+        while True:
+            try:
+                drained.append(self._inbox.get_nowait())
+            except _q.Empty:
+                return drained
+
+    def _loop(self) -> None:
+        # lockstep replicas can never block on the wake event — every rank
+        # must reach the admit-broadcast each iteration
+        while self._running and not self.stop_seen:
+            self.step()
+
+    def _finish(self, hm, done) -> None:
+        if not self.is_leader:
+            for seq in done:
+                if seq in hm.active:
+                    hm.active.remove(seq)
+            return
+        super()._finish(hm, done)
+
+
+    def stop(self) -> None:
+        if self.tp.world > 1 and self.is_leader and not self.stop_seen:
+            self.request_stop()
+            if self._thread is None:
+                self.step()
+        super().stop()
+
+
+def serve_tp_replica(engine: TPEngine) -> None:
+    """Replica main loop: lockstep-step until the leader broadcasts stop."""
+    while not engine.stop_seen:
+        engine.step()

@@ -29,6 +29,8 @@ import torch
 
 from ..ops import dispatch as D
 from .config import ModelConfig, get_config, instance_seed
+from ..parallel.tp import (TPContext, shard_cols, shard_gate_up, shard_qkv,
+                           shard_rows)
 
 
 @dataclass
@@ -84,11 +86,26 @@ def _rand(shape, std: float, gen: torch.Generator, device, dtype):
 class LlamaModel:
     def __init__(self, key: str, device: torch.device,
                  dtype: torch.dtype = torch.bfloat16,
-                 cfg: Optional[ModelConfig] = None):
+                 cfg: Optional[ModelConfig] = None,
+                 tp: Optional[TPContext] = None):
         self.key = key
         self.cfg = cfg or get_config(key)
         self.device = device
         self.dtype = dtype
+        self.tp = tp or TPContext(0, 1)
+        if self.tp.world > 1:
+            from dataclasses import replace
+            c = self.cfg
+            assert c.n_heads % self.tp.world == 0 \
+                and c.n_kv_heads % self.tp.world == 0 \
+                and c.intermediate % self.tp.world == 0, \
+                f"{c.name} not divisible by tp={self.tp.world}"
+            # local (per-shard) architecture: forward + KV cache use this
+            self.lcfg = replace(c, n_heads=c.n_heads // self.tp.world,
+                                n_kv_heads=c.n_kv_heads // self.tp.world,
+                                intermediate=c.intermediate // self.tp.world)
+        else:
+            self.lcfg = self.cfg
         self._init_weights()
 
     def _init_weights(self) -> None:
@@ -104,23 +121,39 @@ class LlamaModel:
         self.embed = _rand((cfg.vocab_size, cfg.hidden), std, gen, gen_dev, dt).to(dev)
         self.layers: List[Dict[str, torch.Tensor]] = []
         for _ in range(cfg.n_layers):
+            tp = self.tp
+            wqkv = _rand((cfg.hidden, cfg.qkv_dim), std, gen, gen_dev, dt)
+            wo = _rand((cfg.q_dim, cfg.hidden), out_std, gen, gen_dev, dt)
+            if tp.world > 1:
+                wqkv = shard_qkv(wqkv, cfg.n_heads, cfg.n_kv_heads,
+                                 cfg.head_dim, tp.rank, tp.world)
+                wo = shard_rows(wo, tp.rank, tp.world)
             layer = {
                 "attn_norm": torch.ones(cfg.hidden, dtype=dt, device=dev),
-                "wqkv": _rand((cfg.hidden, cfg.qkv_dim), std, gen, gen_dev, dt).to(dev),
-                "wo": _rand((cfg.q_dim, cfg.hidden), out_std, gen, gen_dev, dt).to(dev),
+                "wqkv": wqkv.to(dev),
+                "wo": wo.to(dev),
                 "ffn_norm": torch.ones(cfg.hidden, dtype=dt, device=dev),
             }
             if cfg.is_moe:
                 layer["router"] = _rand((cfg.hidden, cfg.n_experts), std, gen, gen_dev, dt).to(dev)
-                layer["w_gate_up"] = _rand(
-                    (cfg.n_experts, cfg.hidden, 2 * cfg.intermediate), std, gen, gen_dev, dt).to(dev)
-                layer["w_down"] = _rand(
-                    (cfg.n_experts, cfg.intermediate, cfg.hidden), out_std, gen, gen_dev, dt).to(dev)
+                wgu = _rand((cfg.n_experts, cfg.hidden, 2 * cfg.intermediate),
+                            std, gen, gen_dev, dt)
+                wdn = _rand((cfg.n_experts, cfg.intermediate, cfg.hidden),
+                            out_std, gen, gen_dev, dt)
+                if tp.world > 1:
+                    wgu = shard_gate_up(wgu, cfg.intermediate, tp.rank, tp.world)
+                    step = cfg.intermediate // tp.world
+                    wdn = wdn[:, tp.rank * step:(tp.rank + 1) * step, :].contiguous()
+                layer["w_gate_up"] = wgu.to(dev)
+                layer["w_down"] = wdn.to(dev)
             else:
-                layer["w_gate_up"] = _rand(
-                    (cfg.hidden, 2 * cfg.intermediate), std, gen, gen_dev, dt).to(dev)
-                layer["w_down"] = _rand(
-                    (cfg.intermediate, cfg.hidden), out_std, gen, gen_dev, dt).to(dev)
+                wgu = _rand((cfg.hidden, 2 * cfg.intermediate), std, gen, gen_dev, dt)
+                wdn = _rand((cfg.intermediate, cfg.hidden), out_std, gen, gen_dev, dt)
+                if tp.world > 1:
+                    wgu = shard_gate_up(wgu, cfg.intermediate, tp.rank, tp.world)
+                    wdn = shard_rows(wdn, tp.rank, tp.world)
+                layer["w_gate_up"] = wgu.to(dev)
+                layer["w_down"] = wdn.to(dev)
             self.layers.append(layer)
         self.final_norm = torch.ones(cfg.hidden, dtype=dt, device=dev)
         if cfg.tie_embeddings:
@@ -138,12 +171,12 @@ class LlamaModel:
         return n * self.embed.element_size()
 
     def new_kv_cache(self, num_blocks: int, block_size: int) -> KVCache:
-        return KVCache(self.cfg, num_blocks, block_size, self.device)
+        return KVCache(self.lcfg, num_blocks, block_size, self.device)
 
     # -- forward -------------------------------------------------------------
 
     def forward(self, batch: ForwardBatch, kv: KVCache) -> torch.Tensor:
-        cfg = self.cfg
+        cfg = self.lcfg     # per-shard head/intermediate dims under TP
         T = batch.total_tokens
         res = torch.empty((T, cfg.hidden), dtype=self.dtype, device=self.device)
         D.gather_rows(res, self.embed, batch.tokens)
@@ -174,6 +207,7 @@ class LlamaModel:
                     batch.tile_q0, batch.tile_qn, batch.tile_seq,
                     batch.tile_pos0, self.scale, max_kv=batch.max_kv)
             proj = attn_out.view(T, cfg.q_dim) @ layer["wo"]
+            self.tp.all_reduce_(proj)
             D.rmsnorm(h, proj, res, layer["ffn_norm"], cfg.rmsnorm_eps)
             if cfg.is_moe:
                 ffn = self._moe_ffn(h, layer)
@@ -183,13 +217,14 @@ class LlamaModel:
                                   device=self.device)
                 D.swiglu(act, gu)
                 ffn = act @ layer["w_down"]
+            self.tp.all_reduce_(ffn)
             next_norm = (self.layers[li + 1]["attn_norm"]
                          if li + 1 < cfg.n_layers else self.final_norm)
             D.rmsnorm(h, ffn, res, next_norm, cfg.rmsnorm_eps)
         return h        # final-normed hidden states [T, hidden]
 
     def _moe_ffn(self, h: torch.Tensor, layer: Dict[str, torch.Tensor]) -> torch.Tensor:
-        cfg = self.cfg
+        cfg = self.lcfg     # local expert intermediate under TP
         logits = (h @ layer["router"]).float()                 # [T, E]
         weights, experts = torch.topk(torch.softmax(logits, dim=-1),
                                       cfg.top_k_experts, dim=-1)
