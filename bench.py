@@ -100,7 +100,7 @@ def main():
     local_keys = rank_model_keys(args.model, args.pool_size, rank)
     engine = LocalEngine(
         local_keys, device=device, kv_gb_per_model=args.kv_gb,
-        embed_model_key="embed-small" if rank == 0 else None)
+        embed_model_key="embed-small")   # every rank joins the RCCL vote
     print(f"[bench t={time.perf_counter() - T_START:.1f}s] rank {rank}: "
           f"engine up, models {local_keys} on {device}",
           file=sys.stderr, flush=True)
@@ -130,7 +130,12 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
     pool = EnginePool(embedder=engine)
     if world > 1:
         from quoracle_amd.parallel.control import ControlClient, RemoteEngine
+        from quoracle_amd.parallel.vote import DistributedEmbedder
         client = ControlClient(list(range(1, world)))
+        # consensus vote embeddings: spread over all ranks, merged by ONE
+        # RCCL all-gather over xGMI
+        pool = EnginePool(embedder=DistributedEmbedder(engine, client, world))
+        pool._embedder_engine = engine
     for r in range(world):
         for key in rank_model_keys(args.model, args.pool_size, r):
             if r == 0:

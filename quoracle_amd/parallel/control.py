@@ -137,6 +137,13 @@ class ControlClient:
             except Exception:
                 pass
 
+    def embed_gather(self, shards, counts) -> None:
+        """Fan the per-rank text shards to the servers; every rank then
+        joins one RCCL all-gather of the embedding blocks (parallel/vote.py)."""
+        for rank, lock in self._locks.items():
+            _send_obj({"kind": "embed_gather", "texts": shards[rank],
+                       "counts": counts}, rank, REQ_TAG, lock)
+
     def barrier_all(self) -> None:
         """Device-synchronized barrier across orchestrator + all servers
         (brackets the bench's timed region)."""
@@ -179,6 +186,11 @@ def serve_engine(engine, orchestrator_rank: int = 0) -> None:
                 torch.cuda.synchronize()
             dist.barrier()
             barrier_times.append(time.perf_counter())
+            continue
+        if msg.get("kind") == "embed_gather":
+            from .vote import all_gather_vote, compute_local_embeddings
+            local = compute_local_embeddings(engine, msg["texts"])
+            all_gather_vote(local, msg["counts"])
             continue
         if msg.get("kind") == "reduce_max":
             elapsed = (barrier_times[-1] - barrier_times[-2]
