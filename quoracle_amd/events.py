@@ -65,6 +65,14 @@ class EventBus:
     def broadcast(self, topic: str, type_: str, payload: Dict[str, Any]) -> Event:
         event = Event(topic=topic, type=type_, payload=payload)
         self._record_history(event)
+        # "*" subscribers get the firehose (monitor UI event stream)
+        for queue in list(self._queues.get("*", ())) if topic != "*" else []:
+            try:
+                queue.put_nowait(event)
+            except asyncio.QueueFull:
+                pass
+        for cb in list(self._callbacks.get("*", ())) if topic != "*" else []:
+            cb(event)
         for queue in list(self._queues.get(topic, ())):
             try:
                 queue.put_nowait(event)

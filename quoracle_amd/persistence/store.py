@@ -270,6 +270,33 @@ class Store:
             f" WHERE agent_id IN ({marks})", tuple(agent_ids))
         return float(rows[0]["total"])
 
+    def cost_rollup(self, agent_id: str) -> Dict[str, Any]:
+        """Recursive descendant cost aggregation via CTE over the agents
+        table's parent links (reference: costs/aggregator.ex:122-163)."""
+        rows = self._query(
+            """
+            WITH RECURSIVE subtree(id) AS (
+                SELECT agent_id FROM agents WHERE agent_id = ?
+                UNION ALL
+                SELECT a.agent_id FROM agents a
+                JOIN subtree s ON a.parent_id = s.id
+            )
+            SELECT c.model, c.category,
+                   COALESCE(SUM(c.amount), 0) AS amount,
+                   COUNT(*) AS entries
+            FROM agent_costs c JOIN subtree s ON c.agent_id = s.id
+            GROUP BY c.model, c.category
+            """, (agent_id,))
+        by_model: Dict[str, float] = {}
+        total = 0.0
+        for r in rows:
+            total += r["amount"]
+            key = r["model"] or r["category"]
+            by_model[key] = by_model.get(key, 0.0) + r["amount"]
+        own = self.total_cost([agent_id])
+        return {"agent_id": agent_id, "total": total, "own": own,
+                "descendants": total - own, "by_model": by_model}
+
     # -- secrets -----------------------------------------------------------------
     def save_secret(self, name: str, value: bytes, description: str = "") -> None:
         self._exec(

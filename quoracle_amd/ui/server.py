@@ -1,0 +1,263 @@
+"""Web monitor: REST + SSE dashboard over the orchestrator.
+
+The MI355X-native replacement for the reference's Phoenix LiveView layer
+(reference: lib/quoracle_web/ — DashboardLive 3-panel task tree / log
+viewer / mailbox, SecretManagementLive tabs, health endpoint, router.ex).
+Instead of LiveView websockets, a single-page dashboard polls the JSON API
+and tails one Server-Sent-Events stream fed by the EventBus (events.py —
+the PubSub equivalent), so everything the reference broadcasts is visible
+here too.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+from typing import Any, Dict, Optional
+
+from fastapi import FastAPI, HTTPException
+from fastapi.responses import HTMLResponse, StreamingResponse
+from pydantic import BaseModel
+
+
+class CreateTaskBody(BaseModel):
+    prompt: str
+    profile: str = "default"
+    budget_limit: Optional[float] = None
+    global_context: Optional[str] = None
+    role: Optional[str] = None
+
+
+class MessageBody(BaseModel):
+    content: str
+    target_agent: Optional[str] = None
+
+
+class SecretBody(BaseModel):
+    name: str
+    value: str
+    description: str = ""
+
+
+class ProfileBody(BaseModel):
+    name: str
+    description: str = ""
+    model_pool: list
+    capability_groups: list = []
+    max_refinement_rounds: int = 4
+    force_reflection: bool = False
+
+
+def create_app(manager) -> FastAPI:
+    runtime = manager.runtime
+    app = FastAPI(title="quoracle-amd monitor")
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok", "agents": len(runtime.registry.all_ids())}
+
+    # -- tasks ---------------------------------------------------------------
+
+    @app.get("/api/tasks")
+    def tasks():
+        return runtime.store.list_tasks()
+
+    @app.post("/api/tasks")
+    async def create_task(body: CreateTaskBody):
+        from ..tasks.manager import TaskError
+        try:
+            return await manager.create_task(
+                body.prompt, body.profile, budget_limit=body.budget_limit,
+                global_context=body.global_context, role=body.role)
+        except TaskError as exc:
+            raise HTTPException(400, exc.reason)
+
+    @app.get("/api/tasks/{task_id}/tree")
+    def task_tree(task_id: str):
+        agents = runtime.store.agents_for_task(task_id)
+        nodes = []
+        for a in agents:
+            entry = runtime.registry.lookup(a["agent_id"])
+            status = entry.actor.state.status if entry else a.get("status")
+            nodes.append({"agent_id": a["agent_id"],
+                          "parent_id": a.get("parent_id"),
+                          "status": status,
+                          "alive": entry is not None})
+        return {"task_id": task_id, "agents": nodes}
+
+    @app.get("/api/tasks/{task_id}/messages")
+    def task_messages(task_id: str):
+        return runtime.store.messages_for_task(task_id)
+
+    @app.post("/api/tasks/{task_id}/message")
+    async def send_message(task_id: str, body: MessageBody):
+        from ..tasks.manager import TaskError
+        try:
+            await manager.send_user_message(task_id, body.content,
+                                            agent_id=body.target_agent)
+        except TaskError as exc:
+            raise HTTPException(400, exc.reason)
+        return {"ok": True}
+
+    @app.post("/api/tasks/{task_id}/pause")
+    async def pause(task_id: str):
+        await manager.pause_task(task_id)
+        return {"ok": True}
+
+    @app.post("/api/tasks/{task_id}/restore")
+    async def restore(task_id: str):
+        return await manager.restore_task(task_id)
+
+    @app.delete("/api/tasks/{task_id}")
+    async def delete(task_id: str):
+        await manager.delete_task(task_id)
+        return {"ok": True}
+
+    # -- agents --------------------------------------------------------------
+
+    @app.get("/api/agents/{agent_id}/logs")
+    def agent_logs(agent_id: str, limit: int = 100):
+        return runtime.store.logs_for_agent(agent_id, limit=limit)
+
+    @app.get("/api/agents/{agent_id}/costs")
+    def agent_costs(agent_id: str):
+        return runtime.store.cost_rollup(agent_id)
+
+    @app.get("/api/agents/{agent_id}/state")
+    def agent_state(agent_id: str):
+        entry = runtime.registry.lookup(agent_id)
+        if entry is None:
+            row = runtime.store.get_agent(agent_id)
+            if row is None:
+                raise HTTPException(404, "unknown agent")
+            return {"agent_id": agent_id, "alive": False,
+                    "status": row.get("status")}
+        st = entry.actor.state
+        return {"agent_id": agent_id, "alive": True, "status": st.status,
+                "model_pool": st.model_pool, "todos": st.todos,
+                "children": list(st.children),
+                "pending_actions": list(st.pending_actions),
+                "history_lengths": {m: len(h) for m, h in
+                                    st.model_histories.items()}}
+
+    # -- config surfaces (SecretManagementLive equivalents) ------------------
+
+    @app.get("/api/profiles")
+    def profiles():
+        return runtime.store.list_profiles()
+
+    @app.post("/api/profiles")
+    def put_profile(body: ProfileBody):
+        from ..governance.profiles import Profile
+        runtime.profiles.put(Profile(
+            name=body.name, description=body.description,
+            model_pool=body.model_pool,
+            capability_groups=body.capability_groups,
+            max_refinement_rounds=body.max_refinement_rounds,
+            force_reflection=body.force_reflection))
+        return {"ok": True}
+
+    @app.get("/api/secrets")
+    def secrets():
+        return {"names": runtime.vault.names()}
+
+    @app.post("/api/secrets")
+    def put_secret(body: SecretBody):
+        runtime.vault.put(body.name, body.value, body.description)
+        return {"ok": True}
+
+    @app.get("/api/engine/stats")
+    def engine_stats():
+        stats = {}
+        embedder = getattr(runtime.engines, "_embedder", None)
+        if embedder is not None and hasattr(embedder, "stats"):
+            stats = dict(embedder.stats)
+        return stats
+
+    # -- event stream (PubSub tail) ------------------------------------------
+
+    @app.get("/api/events")
+    async def events():
+        queue = runtime.bus.subscribe("*", maxsize=500)
+
+        async def stream():
+            try:
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(queue.get(), timeout=15)
+                        data = json.dumps({"topic": ev.topic, "type": ev.type,
+                                           "payload": ev.payload,
+                                           "ts": ev.ts}, default=str)
+                        yield f"data: {data}\n\n"
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                runtime.bus.unsubscribe("*", queue)
+
+        return StreamingResponse(stream(), media_type="text/event-stream")
+
+    @app.get("/", response_class=HTMLResponse)
+    def dashboard():
+        return _DASHBOARD_HTML
+
+    return app
+
+
+_DASHBOARD_HTML = """<!doctype html>
+<html><head><title>quoracle-amd</title><style>
+body{font-family:monospace;background:#111;color:#ddd;margin:0;display:grid;
+grid-template-columns:300px 1fr 1fr;height:100vh}
+.panel{overflow:auto;border-right:1px solid #333;padding:8px}
+h2{font-size:13px;color:#7af;margin:4px 0}
+.agent{margin-left:12px;cursor:pointer}.agent.busy{color:#fc6}
+.agent.waiting{color:#6cf}.agent.dead{color:#666}
+.log{font-size:11px;border-bottom:1px solid #222;padding:2px}
+.log.error{color:#f66}.log.warning{color:#fc6}
+.msg{font-size:11px;padding:3px;border-bottom:1px solid #222}
+textarea,input,select{width:95%;background:#222;color:#ddd;border:1px solid #444}
+button{background:#247;color:#fff;border:0;padding:4px 10px;cursor:pointer}
+</style></head><body>
+<div class=panel id=left><h2>tasks</h2><div id=tasks></div>
+<h2>new task</h2><textarea id=prompt rows=3></textarea>
+<input id=profile value=default placeholder=profile>
+<button onclick=createTask()>create</button></div>
+<div class=panel><h2>agent logs <span id=sel></span></h2><div id=logs></div></div>
+<div class=panel><h2>mailbox / events</h2>
+<textarea id=usermsg rows=2 placeholder="message to task"></textarea>
+<button onclick=sendMsg()>send</button><div id=events></div></div>
+<script>
+let selTask=null, selAgent=null;
+async function j(u,opt){const r=await fetch(u,opt);return r.json()}
+async function refresh(){
+ const tasks=await j('/api/tasks');
+ let html='';
+ for(const t of tasks){
+  html+=`<div><b onclick="selTask='${t.task_id}'">[${t.status}] ${t.task_id}</b>`;
+  const tree=await j(`/api/tasks/${t.task_id}/tree`);
+  for(const a of tree.agents)
+   html+=`<div class="agent ${a.alive?a.status:'dead'}"
+     onclick="pick('${t.task_id}','${a.agent_id}')">${a.agent_id} (${a.status})</div>`;
+  html+='</div>'}
+ document.getElementById('tasks').innerHTML=html;
+ if(selAgent){
+  const logs=await j(`/api/agents/${selAgent}/logs`);
+  document.getElementById('logs').innerHTML=logs.map(l=>
+   `<div class="log ${l.level}">${l.event_type||''} ${l.message||''}</div>`).join('');
+  document.getElementById('sel').textContent=selAgent}}
+function pick(t,a){selTask=t;selAgent=a;refresh()}
+async function createTask(){await j('/api/tasks',{method:'POST',
+ headers:{'Content-Type':'application/json'},
+ body:JSON.stringify({prompt:document.getElementById('prompt').value,
+ profile:document.getElementById('profile').value})});refresh()}
+async function sendMsg(){if(!selTask)return;
+ await j(`/api/tasks/${selTask}/message`,{method:'POST',
+ headers:{'Content-Type':'application/json'},
+ body:JSON.stringify({content:document.getElementById('usermsg').value})})}
+const es=new EventSource('/api/events');
+es.onmessage=e=>{const d=JSON.parse(e.data);
+ const el=document.getElementById('events');
+ el.insertAdjacentHTML('afterbegin',
+  `<div class=msg>[${d.type}] ${d.topic} ${JSON.stringify(d.payload).slice(0,200)}</div>`);
+ while(el.children.length>200)el.removeChild(el.lastChild)};
+setInterval(refresh,2000);refresh();
+</script></body></html>"""

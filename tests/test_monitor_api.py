@@ -1,0 +1,64 @@
+"""Web monitor API (L7 equivalent) over an isolated runtime + FakeEngine."""
+
+import pytest
+
+pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient
+
+from helpers import IDLE, action_json, make_manager
+from quoracle_amd.engine.fake import FakeEngine
+from quoracle_amd.ui.server import create_app
+
+
+@pytest.fixture()
+def client():
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    app = create_app(manager)
+    with TestClient(app) as c:
+        yield c, runtime
+
+
+def test_health_and_task_lifecycle(client):
+    c, runtime = client
+    assert c.get("/health").json()["status"] == "ok"
+    r = c.post("/api/tasks", json={"prompt": "do it", "profile": "default"})
+    assert r.status_code == 200
+    task_id = r.json()["task_id"]
+    root = r.json()["root_agent_id"]
+
+    tasks = c.get("/api/tasks").json()
+    assert any(t["task_id"] == task_id for t in tasks)
+    tree = c.get(f"/api/tasks/{task_id}/tree").json()
+    assert any(a["agent_id"] == root for a in tree["agents"])
+    state = c.get(f"/api/agents/{root}/state").json()
+    assert state["alive"] and state["model_pool"]
+    costs = c.get(f"/api/agents/{root}/costs").json()
+    assert "total" in costs and "descendants" in costs
+    assert c.get(f"/api/agents/{root}/logs").status_code == 200
+    assert c.post(f"/api/tasks/{task_id}/message",
+                  json={"content": "status?"}).json()["ok"]
+    assert c.post(f"/api/tasks/{task_id}/pause").json()["ok"]
+
+
+def test_unknown_profile_rejected(client):
+    c, _ = client
+    r = c.post("/api/tasks", json={"prompt": "x", "profile": "nope"})
+    assert r.status_code == 400
+
+
+def test_profiles_and_secrets(client):
+    c, _ = client
+    r = c.post("/api/profiles", json={
+        "name": "p2", "model_pool": ["fake-a"], "description": "d"})
+    assert r.json()["ok"]
+    assert any(p["name"] == "p2" for p in c.get("/api/profiles").json())
+    assert c.post("/api/secrets", json={
+        "name": "tok", "value": "s3cr3tvalue"}).json()["ok"]
+    assert "tok" in c.get("/api/secrets").json()["names"]
+
+
+def test_dashboard_served(client):
+    c, _ = client
+    r = c.get("/")
+    assert r.status_code == 200 and "quoracle" in r.text
