@@ -101,6 +101,18 @@ class AgentActor:
                 grove["path"], grove.get("skills_path") or "skills")
         return SkillLoader(self.runtime.config.skills_dir, grove_skills)
 
+    async def switch_model_pool(self, new_pool) -> Dict[str, str]:
+        """Runtime model-pool switching with history transfer
+        (reference: agent/history_transfer.ex:38-240)."""
+        from . import history_transfer
+        report = await history_transfer.transfer_histories(
+            self.state, list(new_pool), self.runtime.engines.engine_for,
+            embed_many=self._embed_many())
+        self._persist()
+        self.runtime.bus.log(self.state.agent_id, "info",
+                             "model pool switched", {"report": report})
+        return report
+
     def invalidate_system_prompt(self) -> None:
         self.state.cached_system_prompt = None
 

@@ -240,3 +240,32 @@ async def test_boot_revival():
     assert root_id in results[task_id]["restored"]
     assert runtime2.registry.alive(root_id)
     await manager2.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_history_transfer_on_pool_switch():
+    """Pool switch seeds new models from the largest source history and
+    drops removed models (reference: agent/history_transfer.ex)."""
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("transfer me", "default")
+    root = runtime.registry.lookup(result["root_agent_id"]).actor
+
+    from quoracle_amd.agent.state import history_entry
+    for i in range(5):
+        root.state.append_history(history_entry("event", f"note {i}"),
+                                  models=["fake-a"])
+    root.state.context_lessons["fake-a"] = [{"content": "lesson", "confidence": 1}]
+
+    report = await root.switch_model_pool(["fake-a", "fake-x"])
+    assert report["fake-a"] == "kept"
+    assert report["fake-x"].startswith("seeded_from:fake-a")
+    assert root.state.model_pool == ["fake-a", "fake-x"]
+    assert "fake-b" not in root.state.model_histories
+    assert len(root.state.model_histories["fake-x"]) >= 5
+    assert root.state.context_lessons["fake-x"] == \
+        root.state.context_lessons["fake-a"]
+    # deep copies: mutating the clone must not touch the source
+    root.state.model_histories["fake-x"][0]["content"] = "mutated"
+    assert root.state.model_histories["fake-a"][0]["content"] != "mutated"
+    await manager.supervisor.terminate_tree(root.state.agent_id)
