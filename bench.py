@@ -82,7 +82,7 @@ def main():
         from quoracle_amd.engine.tp_engine import TPEngine, serve_tp_replica
         from quoracle_amd.parallel.tp import TPContext
         tp = TPContext(rank, world)
-        keys = [f"{m}#tp" for m in args.model.split("+")]
+        keys = [f"{m}#tp{i}" for i, m in enumerate(args.model.split("+"))]
         engine = TPEngine(keys, tp, device=device,
                           kv_gb_per_model=args.kv_gb,
                           embed_model_key="embed-small" if rank == 0 else None)
@@ -94,6 +94,7 @@ def main():
         result = asyncio.run(orchestrate(args, engine, device, 1,
                                          pool_keys=keys,
                                          parallelism=f"tp{world}"))
+        result["n_gpus"] = world
         print(json.dumps(result), flush=True)
         return
 

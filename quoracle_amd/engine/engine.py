@@ -233,7 +233,8 @@ class LocalEngine(Engine):
 
     # -- request intake ------------------------------------------------------
 
-    def _submit(self, request: GenerateRequest, completion) -> Optional[GenerateResult]:
+    def _submit(self, request: GenerateRequest, completion,
+                enqueue: bool = True) -> Optional[GenerateResult]:
         hm = self.models.get(request.model_key)
         if hm is None:
             return GenerateResult(model_key=request.model_key,
@@ -265,6 +266,9 @@ class LocalEngine(Engine):
                    params=params, grammar=grammar, generator=gen,
                    _complete=completion)
         seq._session_id = sid                                     # type: ignore
+        if not enqueue:
+            self._made_seq = seq
+            return None
         self._inbox.put(seq)
         self._wake.set()
         return None
@@ -317,17 +321,22 @@ class LocalEngine(Engine):
         self._finish(hm, crashed)
 
     def _admit(self) -> None:
+        if getattr(self, "_tp_direct", False):
+            return      # lockstep engines admit via the admit-broadcast only
         while True:
             try:
                 seq = self._inbox.get_nowait()
             except queue.Empty:
                 return
-            hm = self.models[seq.request.model_key]
-            sess = hm.sessions.get_or_create(seq._session_id)   # type: ignore
-            hit = hm.sessions.match_prefix(sess, seq.prompt)
-            self.stats["prefix_hit_tokens"] += hit
-            seq.session = sess
-            hm.active.append(seq)
+            self._admit_seq(seq)
+
+    def _admit_seq(self, seq: "_Seq") -> None:
+        hm = self.models[seq.request.model_key]
+        sess = hm.sessions.get_or_create(seq._session_id)   # type: ignore
+        hit = hm.sessions.match_prefix(sess, seq.prompt)
+        self.stats["prefix_hit_tokens"] += hit
+        seq.session = sess
+        hm.active.append(seq)
 
     def _expand_forced(self, hm: _HostedModel, seq: _Seq) -> None:
         """Append the grammar's next run of FORCED tokens in one go: known

@@ -47,6 +47,11 @@ class TPEngine(LocalEngine):
         self._idle_steps = 0
         kw.setdefault("tp", tp)
         super().__init__(model_keys, device=device, **kw)
+        # lockstep: batches are composed ONLY from the admit-broadcast —
+        # requests landing between broadcast and admit wait for the next
+        # step (a leader admitting unannounced work would desync the
+        # per-layer all-reduce counts and deadlock the group)
+        self._tp_direct = True
 
     def request_stop(self) -> None:
         """Leader: broadcast shutdown to replicas on the next step."""
@@ -62,7 +67,7 @@ class TPEngine(LocalEngine):
                 dist.broadcast_object_list(box, src=0,
                                            group=self.control_group)
                 for s in seqs:
-                    self._inbox.put(s)
+                    self._admit_seq(s)
                 if stop:
                     self.stop_seen = True
             else:
@@ -73,11 +78,13 @@ class TPEngine(LocalEngine):
                 if stop:
                     self.stop_seen = True
                 for req in reqs:
-                    err = self._submit(req, None)
+                    err = self._submit(req, None, enqueue=False)
                     if err is not None:
                         print(f"[tp-engine r{self.tp.rank}] divergent "
                               f"submit: {err.error}", file=sys.stderr,
                               flush=True)
+                    else:
+                        self._admit_seq(self._made_seq)
         worked = super().step()
         if self.tp.world > 1:
             # deterministic shared idle backoff: every rank sees the same
@@ -115,6 +122,13 @@ class TPEngine(LocalEngine):
             self.request_stop()
             if self._thread is None:
                 self.step()
+            else:
+                # the loop thread must broadcast the stop sentinel BEFORE
+                # _running flips false, or replicas hang in their next
+                # admit-broadcast against a dead leader
+                deadline = time.monotonic() + 30
+                while not self.stop_seen and time.monotonic() < deadline:
+                    time.sleep(0.002)
         super().stop()
 
 
