@@ -204,3 +204,182 @@ paged_attn_prefill_mfma_kernel(
         f2bf(o_acc1[r] / denom);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Context-split MFMA prefill: small chunks (grammar forced-runs) have only
+// a handful of tiles, so (ntiles, Hq) leaves the chip empty while each WG
+// walks the whole cached context.  Partition the walk over NS workgroups;
+// partials land in the SAME layout as the VALU split kernel
+// (part_m/l: [ntiles, Hq, NS, QT], part_acc: [..., D]) so the existing
+// paged_attn_prefill_reduce_kernel combines them.
+// Grid: (ntiles, Hq, NS); block 256 (4 waves).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+paged_attn_prefill_mfma_split_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ, int NS) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int split = blockIdx.z;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+  const int chunks = (kv_limit + MF_KCHUNK - 1) / MF_KCHUNK;
+  const int per = (chunks + NS - 1) / NS;
+  const int c0 = split * per * MF_KCHUNK;
+  const int c1 = min(kv_limit, (split + 1) * per * MF_KCHUNK);
+  const bool dead = (c0 >= kv_limit);
+
+  __shared__ bf16 q_s[MF_QT * KP];
+  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ float s_s[MF_QT * SP];
+  __shared__ bf16 p_s[MF_QT * VP];
+  __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
+
+  for (int i = tid; i < MF_QT * MF_D / 8; i += 256) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
+  if (tid < MF_QT) {
+    m_s[tid] = -INFINITY;
+    l_s[tid] = 0.f;
+  }
+  __syncthreads();
+
+  f32x4_t o_acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t o_acc1 = {0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int a_row = lane & 15;
+  const int a_koff = (lane >> 4) * 8;
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+
+  for (int start = c0; start < c1; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, c1 - start);
+
+    for (int i = tid; i < MF_KCHUNK * MF_D / 8; i += 256) {
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kval = make_uint4(0, 0, 0, 0);
+      uint4 vval = make_uint4(0, 0, 0, 0);
+      if (key < clen) {
+        const int token = start + key;
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off = blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kval = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vval = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kval;
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vval);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+    }
+    __syncthreads();
+
+    {
+      f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            q_s + a_row * KP + kk * 32 + a_koff);
+        bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = c_row0 + r;
+        const int col = wave * 16 + c_col;
+        const int token = start + col;
+        const bool ok = (token <= pos0 + min(row, qn - 1)) && (col < clen);
+        s_s[row * SP + col] = ok ? s_acc[r] * scale : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+    {
+      const int row = tid >> 4;
+      const int sub = tid & 15;
+      float v0 = s_s[row * SP + sub];
+      float v1 = s_s[row * SP + sub + 16];
+      float v2 = s_s[row * SP + sub + 32];
+      float v3 = s_s[row * SP + sub + 48];
+      float mymax = fmaxf(fmaxf(v0, v1), fmaxf(v2, v3));
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        mymax = fmaxf(mymax, __shfl_xor(mymax, w, 16));
+      const float m_old = m_s[row];
+      const float mn = fmaxf(m_old, mymax);
+      const float alpha = (m_old == -INFINITY) ? 0.f : __expf(m_old - mn);
+      float p0 = (v0 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v0 - mn);
+      float p1 = (v1 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v1 - mn);
+      float p2 = (v2 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v2 - mn);
+      float p3 = (v3 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v3 - mn);
+      float psum = p0 + p1 + p2 + p3;
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        psum += __shfl_xor(psum, w, 16);
+      if (sub == 0) {
+        l_s[row] = l_s[row] * alpha + psum;
+        m_s[row] = mn;
+        alpha_s[row] = alpha;
+      }
+      p_s[row * VP + sub] = f2bf(p0);
+      p_s[row * VP + sub + 16] = f2bf(p1);
+      p_s[row * VP + sub + 32] = f2bf(p2);
+      p_s[row * VP + sub + 48] = f2bf(p3);
+    }
+    __syncthreads();
+
+    {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float a = alpha_s[c_row0 + r];
+        o_acc0[r] *= a;
+        o_acc1[r] *= a;
+      }
+#pragma unroll
+      for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            p_s + a_row * VP + kk * 32 + a_koff);
+        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
+        o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // partials out (unnormalized; reduce kernel combines across splits)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = c_row0 + r;
+    if (row >= MF_QT) continue;
+    const long base = (((long)tile * Hq + h) * NS + split) * MF_QT + row;
+    if (c_col == 0 && wave == 0) {
+      part_m[base] = dead ? -INFINITY : m_s[row];
+      part_l[base] = dead ? 0.f : l_s[row];
+    }
+    part_acc[base * MF_D + wave * 32 + c_col] = dead ? 0.f : o_acc0[r];
+    part_acc[base * MF_D + wave * 32 + 16 + c_col] = dead ? 0.f : o_acc1[r];
+  }
+}

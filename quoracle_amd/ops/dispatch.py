@@ -92,9 +92,21 @@ def paged_attn_prefill(out, q, kcache, vcache, block_tables, tile_q0,
         ntiles = tile_q0.shape[0]
         Hq = q.shape[1]
         D = q.shape[2]
-        # big prefills with D=128 go to the MFMA-tiled kernel (matrix cores)
+        # D=128 prefill goes to the matrix cores; small grids additionally
+        # context-split so the KV walk parallelizes across the chip
         ns = max(1, min(16, 2048 // max(1, ntiles * Hq)))
-        if D == 128 and ns == 1 and not os.environ.get("QUORACLE_NO_MFMA_ATTN"):
+        if D == 128 and not os.environ.get("QUORACLE_NO_MFMA_ATTN"):
+            if ns > 1 and max_kv >= 1024:
+                QT = 16
+                part_m = torch.empty((ntiles, Hq, ns, QT),
+                                     dtype=torch.float32, device=q.device)
+                part_l = torch.empty_like(part_m)
+                part_acc = torch.empty((ntiles, Hq, ns, QT, D),
+                                       dtype=torch.float32, device=q.device)
+                ext().paged_attn_prefill_mfma_split(
+                    out, q, kcache, vcache, block_tables, tile_q0, tile_qn,
+                    tile_seq, tile_pos0, scale, part_m, part_l, part_acc)
+                return out
             ext().paged_attn_prefill_mfma(out, q, kcache, vcache,
                                           block_tables, tile_q0, tile_qn,
                                           tile_seq, tile_pos0, scale)

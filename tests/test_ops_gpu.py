@@ -287,3 +287,34 @@ def test_paged_attn_prefill_mfma_matches_reference(dev):
     err = (out.float() - ref).abs().max().item()
     assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
         f"max err {err}"
+
+
+def test_paged_attn_prefill_mfma_split_matches_reference(dev):
+    """Context-split MFMA prefill (small chunk over long cached context)."""
+    ops = _ops()
+    torch.manual_seed(17)
+    Hq, Hkv, D, BS = 32, 8, 128, 16
+    scale = D ** -0.5
+    cached, new = 2000, 23
+    total = cached + new
+    k = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    kcache, vcache, tables, ctx = _build_paged_cache(dev, [(k, v)], Hkv, D, BS)
+    q = torch.randn(new, Hq, D, device=dev, dtype=torch.bfloat16)
+    out = torch.empty_like(q)
+    ntiles = (new + 15) // 16
+    t0 = torch.arange(ntiles, dtype=torch.int32, device=dev) * 16
+    qn = torch.clamp(torch.full_like(t0, new) - t0, max=16)
+    tseq = torch.zeros_like(t0)
+    tpos = t0 + cached
+    NS = 8
+    part_m = torch.empty((ntiles, Hq, NS, 16), dtype=torch.float32, device=dev)
+    part_l = torch.empty_like(part_m)
+    part_acc = torch.empty((ntiles, Hq, NS, 16, D), dtype=torch.float32,
+                           device=dev)
+    ops.ext().paged_attn_prefill_mfma_split(out, q, kcache, vcache, tables,
+                                            t0, qn, tseq, tpos, scale,
+                                            part_m, part_l, part_acc)
+    ref = reference.attention(q, k, v, scale, causal_offset=cached)
+    assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
+        f"max err {(out.float() - ref).abs().max().item()}"
