@@ -236,6 +236,10 @@ def safe_read_file(source_path: str, grove_path: str) -> str:
 _FRONTMATTER_RE = re.compile(r"\A---\s*\n(.*?)\n---\s*(\n|\Z)", re.DOTALL)
 
 BOOTSTRAP_FIELDS = [
+    # inline values and *_file indirections are both accepted; a _file field
+    # resolves to the file's contents at load (path-security checked)
+    "global_context", "task_description", "success_criteria",
+    "immediate_context", "approach_guidance",
     "global_context_file", "task_description_file", "success_criteria_file",
     "immediate_context_file", "approach_guidance_file", "global_constraints",
     "output_style", "role", "cognitive_style", "delegation_strategy",
