@@ -34,10 +34,14 @@ def test_code_sandbox_rules_enforced():
     G.check_shell_command("ls workspace", g["hard_rules"])
     with pytest.raises(G.HardRuleViolation):
         G.check_action("fetch_web", g["hard_rules"])
-    G.check_file_access("workspace/a.py", "write", g["confinement"],
+    base = os.path.join(ROOT, "code-sandbox")
+    G.check_file_access(os.path.join(base, "workspace/a.py"), "write",
+                        g["confinement"],
                         confinement_mode=g["confinement_mode"])
-    G.check_file_access("GROVE.md", "read", g["confinement"],
+    G.check_file_access(os.path.join(base, "GROVE.md"), "read",
+                        g["confinement"],
                         confinement_mode=g["confinement_mode"])
     with pytest.raises(G.ConfinementViolation):
-        G.check_file_access("GROVE.md", "write", g["confinement"],
+        G.check_file_access(os.path.join(base, "GROVE.md"), "write",
+                            g["confinement"],
                             confinement_mode=g["confinement_mode"])
