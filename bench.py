@@ -35,7 +35,7 @@ def parse_args():
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--pool-size", type=int, default=3)
     p.add_argument("--agents-per-gpu", type=int, default=8)
-    p.add_argument("--kv-gb", type=float, default=4.0)
+    p.add_argument("--kv-gb", type=float, default=24.0)
     p.add_argument("--device", default=None, help="override (e.g. cpu)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: the whole job is ONE "

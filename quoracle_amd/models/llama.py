@@ -161,6 +161,35 @@ class LlamaModel:
         else:
             self.lm_head = _rand((cfg.vocab_size, cfg.hidden), std, gen, gen_dev, dt).to(dev)
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self._register_cpp()
+
+    def _register_cpp(self) -> None:
+        """Hand the weights to the C++ forward driver (GPU, non-TP): the
+        whole layer loop then runs in one native call instead of ~400
+        Python op dispatches (ops/csrc/forward.h)."""
+        self.use_cpp = False
+        if self.device.type != "cuda" or self.tp.world > 1:
+            return
+        import os as _os
+        if _os.environ.get("QUORACLE_NO_CPP_FWD"):
+            return
+        from .. import ops as _ops
+        if not _ops.available():
+            return
+        lcfg = self.lcfg
+        layer_lists = []
+        for layer in self.layers:
+            lw = [layer["attn_norm"], layer["wqkv"], layer["wo"],
+                  layer["ffn_norm"], layer["w_gate_up"], layer["w_down"]]
+            if lcfg.is_moe:
+                lw.append(layer["router"])
+            layer_lists.append(lw)
+        _ops.ext().register_model(
+            self.key, self.embed, self.lm_head, self.final_norm, layer_lists,
+            lcfg.n_heads, lcfg.n_kv_heads, lcfg.head_dim, lcfg.intermediate,
+            lcfg.rope_theta, lcfg.rmsnorm_eps, lcfg.is_moe,
+            lcfg.top_k_experts)
+        self.use_cpp = True
 
     def param_bytes(self) -> int:
         n = self.embed.numel() + self.final_norm.numel()
@@ -176,6 +205,15 @@ class LlamaModel:
     # -- forward -------------------------------------------------------------
 
     def forward(self, batch: ForwardBatch, kv: KVCache) -> torch.Tensor:
+        if getattr(self, "use_cpp", False):
+            from .. import ops as _ops
+            import os as _os
+            return _ops.ext().llama_forward(
+                self.key, batch.tokens, batch.positions, batch.slots,
+                batch.block_tables, batch.n_decode, batch.ctx_lens,
+                batch.max_ctx, batch.tile_q0, batch.tile_qn, batch.tile_seq,
+                batch.tile_pos0, batch.max_kv, kv.k, kv.v,
+                bool(_os.environ.get("QUORACLE_NO_MFMA_ATTN")))
         cfg = self.lcfg     # per-shard head/intermediate dims under TP
         T = batch.total_tokens
         res = torch.empty((T, cfg.hidden), dtype=self.dtype, device=self.device)
@@ -248,6 +286,9 @@ class LlamaModel:
     def compute_logits(self, hidden: torch.Tensor,
                        rows: torch.Tensor) -> torch.Tensor:
         """Logits only for the rows that need sampling: [R, vocab] fp32."""
+        if getattr(self, "use_cpp", False):
+            from .. import ops as _ops
+            return _ops.ext().llama_logits(self.key, hidden, rows)
         return (hidden[rows] @ self.lm_head.T).float()
 
     # -- embedding-model path (consensus vote) -------------------------------

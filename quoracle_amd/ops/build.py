@@ -30,7 +30,7 @@ def _torch_paths():
 def build(force: bool = False, verbose: bool = True) -> str:
     sources = [os.path.join(CSRC, "bindings.cpp")]
     headers = [os.path.join(CSRC, f) for f in
-               ("common.h", "attention.hip", "attention_mfma.hip", "elementwise.hip")]
+               ("common.h", "forward.h", "attention.hip", "attention_mfma.hip", "elementwise.hip")]
     if not force and os.path.exists(SO_PATH):
         newest_src = max(os.path.getmtime(p) for p in sources + headers)
         if os.path.getmtime(SO_PATH) >= newest_src:

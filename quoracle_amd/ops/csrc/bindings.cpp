@@ -352,6 +352,8 @@ void gather_rows(torch::Tensor out, torch::Tensor src, torch::Tensor rows) {
   HIP_CHECK_KERNEL();
 }
 
+#include "forward.h"
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -375,4 +377,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cosine_sim_matrix", &cosine_sim_matrix,
         "Pairwise cosine similarity matrix (consensus vote)");
   m.def("gather_rows", &gather_rows, "Embedding row gather (bf16)");
+  m.def("register_model", &qforward::register_model,
+        "Register model weights with the C++ forward driver");
+  m.def("unregister_model", &qforward::unregister_model,
+        "Drop a registered model");
+  m.def("llama_forward", &qforward::forward,
+        "Full transformer forward (layer loop in native code)");
+  m.def("llama_logits", &qforward::compute_logits,
+        "lm_head logits for selected rows");
 }

@@ -91,3 +91,28 @@ def test_gpu_constrained_generate(dev):
     import math
     assert math.isclose(sum(a * b for a, b in zip(v[0], v[1])), 1.0,
                         abs_tol=1e-3)
+
+
+def test_cpp_forward_matches_python(dev):
+    """The native (C++ driver) forward equals the Python-loop forward."""
+    import os
+    from quoracle_amd.models import LlamaModel
+    m_cpp = LlamaModel("tiny#fwd", dev)
+    assert m_cpp.use_cpp, "C++ forward driver not active on GPU"
+    os.environ["QUORACLE_NO_CPP_FWD"] = "1"
+    try:
+        m_py = LlamaModel("tiny#fwd", dev)
+        assert not m_py.use_cpp
+    finally:
+        del os.environ["QUORACLE_NO_CPP_FWD"]
+    batch_a, nb = _mk_batch(dev)
+    batch_b, _ = _mk_batch(dev)
+    kv_a = m_cpp.new_kv_cache(nb, 16)
+    kv_b = m_py.new_kv_cache(nb, 16)
+    h_cpp = m_cpp.forward(batch_a, kv_a)
+    h_py = m_py.forward(batch_b, kv_b)
+    assert torch.allclose(h_cpp.float(), h_py.float(), atol=1e-2, rtol=1e-2)
+    rows = torch.tensor([0, 5], device=dev)
+    l_cpp = m_cpp.compute_logits(h_cpp, rows)
+    l_py = m_py.compute_logits(h_py, rows)
+    assert torch.allclose(l_cpp, l_py, atol=1e-2, rtol=1e-2)
