@@ -319,6 +319,39 @@ class LocalEngine(Engine):
             seq.result = GenerateResult(
                 model_key=hm.key, error=f"engine_error:{exc}")
         self._finish(hm, crashed)
+        # Watchdog: repeated failures mean corrupted
+        # device state — rebuild the model's KV world.  Conversation
+        # history lives in the orchestrator, so sessions re-prefill
+        # lazily via the prefix cache on the next request; the failed
+        # shard simply drops out of votes until then (consensus already
+        # tolerates partial pools).
+        hm.crash_count = getattr(hm, "crash_count", 0) + 1
+        if hm.crash_count >= 2:
+            self.reset_model(hm.key)
+
+    def reset_model(self, key: str) -> None:
+        """Rebuild a hosted model's KV cache, block manager and sessions
+        after a device fault; weights stay resident (they are immutable)."""
+        hm = self.models.get(key)
+        if hm is None:
+            return
+        print(f"[engine {key}] watchdog reset: rebuilding KV state",
+              file=sys.stderr, flush=True)
+        crashed = list(hm.active)
+        for seq in crashed:
+            if seq.result is None:
+                seq.result = GenerateResult(model_key=key,
+                                            error="engine_reset")
+        self._finish(hm, crashed)
+        blocks = hm.mgr.num_blocks
+        bs = hm.kv.block_size
+        for t in hm.kv.k + hm.kv.v:
+            t.zero_()
+        hm.mgr = BlockManager(blocks, bs)
+        hm.sessions = SessionCache(hm.mgr)
+        scratch = hm.mgr.alloc(1)[0]
+        hm.graphs.scratch_block = scratch
+        hm.crash_count = 0
 
     def _admit(self) -> None:
         if getattr(self, "_tp_direct", False):

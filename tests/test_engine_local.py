@@ -132,3 +132,34 @@ def test_embedding_path_cpu():
     import math
     dot_same = sum(a * b for a, b in zip(vecs[0], vecs[1]))
     assert math.isclose(dot_same, 1.0, abs_tol=1e-3)
+
+
+def test_engine_watchdog_resets_model():
+    """Two consecutive step crashes trigger a KV-world rebuild; the engine
+    keeps serving afterwards (SURVEY.md §5.3 failure recovery)."""
+    eng = LocalEngine(["tiny#wd"], device=torch.device("cpu"),
+                      kv_blocks_override=128, embed_model_key=None,
+                      prefill_chunk=64)
+    hm = eng.models["tiny#wd"]
+    r1 = eng.generate_sync(_req(model_key="tiny#wd", session_id="w1"),
+                           timeout=120)
+    assert r1.ok
+    used_before = hm.mgr.free_blocks
+    # inject two crashes
+    boom = RuntimeError("simulated HIP fault")
+    class _Boom:
+        def __getattr__(self, name):
+            raise boom
+    for _ in range(2):
+        eng._submit(_req(model_key="tiny#wd", session_id="w2"),
+                    ("sync", __import__("threading").Event(), []))
+        real_launch = eng._launch_model
+        eng._launch_model = lambda h: (_ for _ in ()).throw(boom)
+        eng.step()
+        eng._launch_model = real_launch
+    # KV world rebuilt: all blocks free again (minus scratch)
+    assert hm.mgr.free_blocks == 127
+    assert hm.crash_count == 0
+    r2 = eng.generate_sync(_req(model_key="tiny#wd", session_id="w3"),
+                           timeout=120)
+    assert r2.ok
