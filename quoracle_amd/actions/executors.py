@@ -667,6 +667,12 @@ async def execute_batch_sync(ctx) -> Dict[str, Any]:
             return {"status": "stopped_on_error", "results": results,
                     "completed": len(results) - 1}
         results.append({"action": spec["action"], "result": result})
+        # executors that report failure as a result payload (file ops,
+        # shell exit codes) also stop the batch (reference: batch_sync.ex
+        # stop-on-first-error)
+        if isinstance(result, dict) and result.get("error"):
+            return {"status": "stopped_on_error", "results": results,
+                    "completed": len(results) - 1}
     return {"status": "completed", "results": results,
             "completed": len(results)}
 
