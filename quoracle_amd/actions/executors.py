@@ -492,8 +492,10 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
 
 async def execute_answer_engine(ctx) -> Dict[str, Any]:
     prompt = ctx.params["prompt"]
-    model = ctx.runtime.extras.get("answer_engine_model") or \
-        (ctx.agent.state.model_pool[0] if ctx.agent.state.model_pool else None)
+    model = (ctx.runtime.config.model_roles.get("answer_engine")
+             or ctx.runtime.extras.get("answer_engine_model")
+             or (ctx.agent.state.model_pool[0]
+                 if ctx.agent.state.model_pool else None))
     if model is None:
         return _err("no_answer_engine_model")
     engine = ctx.runtime.engines.engine_for(model)

@@ -21,6 +21,11 @@ from ..registry import Registry
 
 @dataclass
 class RuntimeConfig:
+    # Role-based model settings (reference: models/config_model_settings.ex):
+    # which hosted model serves each auxiliary role.  Empty entries fall
+    # back per call site (embedding -> pool embedder, answer_engine ->
+    # first pool model, summarization -> the history-owning model itself).
+    model_roles: Dict[str, str] = field(default_factory=dict)
     groves_dir: Optional[str] = None
     skills_dir: Optional[str] = None
     default_working_dir: str = "/tmp"
