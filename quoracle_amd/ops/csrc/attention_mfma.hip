@@ -90,27 +90,49 @@ paged_attn_prefill_mfma_kernel(
   const int c_col = lane & 15;
   const int c_row0 = (lane >> 4) * 4;
 
-  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
-    const int clen = min(MF_KCHUNK, kv_limit - start);
-
-    // ---- stage K (row-major) and V (transposed) ---------------------------
-    for (int i = tid; i < MF_KCHUNK * MF_D / 8; i += 256) {
+  // T14 software pipeline (guide §5.5): chunk c+1's K/V rows ride in
+  // registers while chunk c computes; the LDS write happens after the
+  // barrier, so HBM latency hides under the MFMA phases.
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      uint4 kval = make_uint4(0, 0, 0, 0);
-      uint4 vval = make_uint4(0, 0, 0, 0);
-      if (key < clen) {
-        const int token = start + key;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
         const long blk = bt[(long)seq * MAXB + token / BS];
-        const long off = blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
-        kval = reinterpret_cast<const uint4 *>(kc + off)[0];
-        vval = reinterpret_cast<const uint4 *>(vc + off)[0];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
       }
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kval;
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vval);
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
     }
+  };
+
+  issue_loads(0, kv_limit);
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+
+    // staged K/V (loaded last iteration) -> LDS; prefetch next chunk
+    write_staged();
     __syncthreads();
+    if (start + MF_KCHUNK < kv_limit)
+      issue_loads(start + MF_KCHUNK, kv_limit);
 
     // ---- S = Q·K^T (wave w: key block w*16..w*16+15) ----------------------
     {
@@ -271,26 +293,48 @@ paged_attn_prefill_mfma_split_kernel(
   const int c_col = lane & 15;
   const int c_row0 = (lane >> 4) * 4;
 
-  for (int start = c0; start < c1; start += MF_KCHUNK) {
-    const int clen = min(MF_KCHUNK, c1 - start);
-
-    for (int i = tid; i < MF_KCHUNK * MF_D / 8; i += 256) {
+  // T14 software pipeline (guide §5.5): chunk c+1's K/V rows ride in
+  // registers while chunk c computes; the LDS write happens after the
+  // barrier, so HBM latency hides under the MFMA phases.
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      uint4 kval = make_uint4(0, 0, 0, 0);
-      uint4 vval = make_uint4(0, 0, 0, 0);
-      if (key < clen) {
-        const int token = start + key;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
         const long blk = bt[(long)seq * MAXB + token / BS];
-        const long off = blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
-        kval = reinterpret_cast<const uint4 *>(kc + off)[0];
-        vval = reinterpret_cast<const uint4 *>(vc + off)[0];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
       }
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kval;
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vval);
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
     }
+  };
+
+  issue_loads(c0, c1);
+  for (int start = c0; start < c1; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, c1 - start);
+
+    write_staged();
     __syncthreads();
+    if (start + MF_KCHUNK < c1)
+      issue_loads(start + MF_KCHUNK, c1);
 
     {
       f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
