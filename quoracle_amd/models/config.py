@@ -72,6 +72,13 @@ PRESETS: Dict[str, ModelConfig] = {
                         intermediate=128, rope_theta=10000.0,
                         max_context=32768, max_output=1024,
                         tie_embeddings=True),
+    # CPU-testable MoE toy (engine-level Mixtral-path tests)
+    "tiny-moe": ModelConfig("tiny-moe", vocab_size=512, n_layers=2, hidden=64,
+                            n_heads=4, n_kv_heads=2, head_dim=16,
+                            intermediate=128, rope_theta=10000.0,
+                            max_context=32768, max_output=1024,
+                            tie_embeddings=True, n_experts=4,
+                            top_k_experts=2),
     # GPT-2-small scale, llama-style blocks (BASELINE config 1)
     "gpt2s": ModelConfig("gpt2s", vocab_size=50304, n_layers=12, hidden=768,
                          n_heads=12, n_kv_heads=12, head_dim=64,

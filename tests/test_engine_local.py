@@ -163,3 +163,16 @@ def test_engine_watchdog_resets_model():
     r2 = eng.generate_sync(_req(model_key="tiny#wd", session_id="w3"),
                            timeout=120)
     assert r2.ok
+
+
+def test_moe_model_generates_through_engine():
+    """Mixtral-style MoE path through the full engine (router + per-expert
+    FFN + grammar decode) on CPU."""
+    eng = LocalEngine(["tiny-moe#0"], device=torch.device("cpu"),
+                      kv_blocks_override=256, embed_model_key=None,
+                      prefill_chunk=64)
+    r = eng.generate_sync(_req(model_key="tiny-moe#0", session_id="moe1",
+                               max_tokens=400), timeout=240)
+    assert r.ok, r.error
+    parsed = json.loads(r.text)
+    assert parsed["action"] in {"orient", "send_message", "todo", "wait"}
