@@ -64,8 +64,20 @@ async def reflect(
             continue
         lessons_raw = parsed.get("lessons")
         state = parsed.get("state")
-        lessons = [{"text": l if isinstance(l, str) else json.dumps(l, default=str),
-                    "confidence": 1}
+        def _text(l):
+            if isinstance(l, str):
+                return l
+            if isinstance(l, dict):
+                t = l.get("text") or l.get("content") or l.get("lesson")
+                if isinstance(t, str):
+                    return t
+            return json.dumps(l, default=str)
+
+        def _conf(l):
+            c = l.get("confidence") if isinstance(l, dict) else None
+            return c if isinstance(c, (int, float)) else 1
+
+        lessons = [{"text": _text(l), "confidence": _conf(l)}
                    for l in (lessons_raw or []) if l]
         return lessons, state if isinstance(state, dict) else None
     raise RuntimeError("reflection_failed")

@@ -1,0 +1,103 @@
+"""ACE condensation behavior: 80%-oldest eviction with
+reflection lessons, model-initiated condense:N, fallback artifact, lesson
+dedup by cosine, proactive ensure_fits loop."""
+
+import json
+
+import pytest
+
+from quoracle_amd.agent import condensation as cond
+from quoracle_amd.agent.lessons import merge_lessons
+from quoracle_amd.agent.state import AgentState, history_entry
+from quoracle_amd.engine.fake import FakeEngine, deterministic_embedding
+
+
+def _state(model="fake-a", entries=12):
+    st = AgentState(agent_id="a", task_id="t", parent_id=None,
+                    profile="p", model_pool=[model], capability_groups=[])
+    st.init_model_maps()
+    for i in range(entries):
+        st.append_history(history_entry("event", f"step {i} " + "x" * 50))
+    return st
+
+
+def _reflective_engine():
+    # reflector asks the SAME model to extract lessons/state as JSON
+    return FakeEngine(default_response=json.dumps({
+        "lessons": [{"text": "always check the logs", "confidence": 1}],
+        "state": {"progress": "mid-task"}}))
+
+
+@pytest.mark.asyncio
+async def test_condensation_evicts_and_extracts_lessons():
+    st = _state()
+    eng = _reflective_engine()
+    before = len(st.model_histories["fake-a"])
+    ok = await cond.condense_model_history(st, "fake-a", eng)
+    assert ok
+    after = st.model_histories["fake-a"]
+    assert len(after) < before
+    # eviction marker appended at the oldest end
+    assert "condensed" in after[-1]["content"]
+    lessons = st.context_lessons["fake-a"]
+    assert lessons and lessons[0]["text"] == "always check the logs"
+    assert st.model_states["fake-a"] == {"progress": "mid-task"}
+
+
+@pytest.mark.asyncio
+async def test_model_initiated_condense_n():
+    st = _state(entries=10)
+    eng = _reflective_engine()
+    await cond.condense_model_history(st, "fake-a", eng, n_oldest=4)
+    # 10 - 4 evicted + 1 marker
+    assert len(st.model_histories["fake-a"]) == 7
+
+
+@pytest.mark.asyncio
+async def test_reflection_failure_keeps_fallback_artifact():
+    st = _state()
+    eng = FakeEngine(default_response="not json at all {{{")
+    eng.fail_model("fake-a")
+    ok = await cond.condense_model_history(st, "fake-a", eng)
+    assert ok
+    tail = st.model_histories["fake-a"][-1]["content"]
+    assert "fallback artifact" in tail
+    assert "step 0" in tail          # the lost content survives truncated
+
+
+@pytest.mark.asyncio
+async def test_ensure_fits_condenses_until_output_floor():
+    st = _state(entries=30)
+    eng = _reflective_engine()
+    # context limit so small that the projected output budget is below the
+    # 4096-token floor until history shrinks
+    eng._context_limits["fake-a"] = 4300
+    calls = {"n": 0}
+
+    def fake_input_tokens():
+        calls["n"] += 1
+        h = st.model_histories["fake-a"]
+        return sum(eng.count_tokens(str(e.get("content", ""))) for e in h) + 100
+
+    await cond.ensure_fits(st, "fake-a", eng, fake_input_tokens)
+    assert calls["n"] >= 2           # at least one condensation pass ran
+    assert len(st.model_histories["fake-a"]) < 30
+
+
+def test_lesson_dedup_cosine_and_prune():
+    embed = lambda texts: [deterministic_embedding(t) for t in texts]
+    base = [{"text": "always check the logs first", "confidence": 1}]
+    merged = merge_lessons(base,
+                           [{"text": "always check the logs first",
+                             "confidence": 1}], embed)
+    # near-identical lesson merges: confidence bumped, no duplicate
+    assert len(merged) == 1 and merged[0]["confidence"] >= 2
+    merged = merge_lessons(merged,
+                           [{"text": "entirely unrelated topic about gpus",
+                             "confidence": 1}], embed)
+    assert len(merged) == 2
+    # prune keeps at most 100
+    many = [{"text": f"unique lesson number {i} {'z' * i}",
+             "confidence": 1} for i in range(130)]
+    pruned = merge_lessons([], many, embed)
+    assert len(pruned) <= 100
