@@ -141,6 +141,16 @@ async def execute_action(ctx: ActionContext) -> Dict[str, Any]:
     result = security_mod.scrub_output(result, secret_values)
     result = security_mod.wrap_untrusted_result(ctx.action, result)
 
+    # 7b. Image payloads become on-disk artifacts + placeholders so binary
+    # blobs never enter model histories (reference: agent/image_detector.ex;
+    # compression is a documented divergence — no vision model hosted)
+    from ..utils import images as images_mod
+    result, image_artifacts = images_mod.extract_images(result)
+    if image_artifacts:
+        runtime.bus.log(state.agent_id, "info",
+                        f"{len(image_artifacts)} image artifact(s) detected",
+                        {"artifacts": image_artifacts})
+
     # 8. Broadcast + persist
     runtime.bus.action_event(state.agent_id, "completed", ctx.action, ctx.action_id,
                              {"elapsed_ms": elapsed_ms})

@@ -142,3 +142,23 @@ async def test_hard_rule_blocks_shell_pattern():
     with pytest.raises(R.ActionError):
         await R.execute_action(_ctx(actor, runtime, "execute_shell",
                                     {"command": "curl http://x"}))
+
+
+@pytest.mark.asyncio
+async def test_image_results_become_artifacts(tmp_path, monkeypatch):
+    """Binary image payloads in action results are stored as artifacts and
+    replaced with placeholders (reference: image_detector.ex)."""
+    monkeypatch.setenv("QUORACLE_IMAGE_DIR", str(tmp_path))
+    runtime = make_runtime()
+    actor = _actor(runtime)
+    png = b"\x89PNG\r\n\x1a\n" + b"\x00" * 64
+    import base64
+    data_url = "data:image/png;base64," + base64.b64encode(png).decode()
+    src = tmp_path / "page.txt"
+    src.write_text("before " + data_url + " after")
+    res = await R.execute_action(_ctx(actor, runtime, "file_read",
+                                      {"path": str(src)}))
+    assert "image artifact" in res["content"]
+    assert "base64" not in res["content"]
+    stored = list(tmp_path.glob("*.png"))
+    assert stored and stored[0].read_bytes() == png
