@@ -28,7 +28,7 @@ _MAGIC = [
 ]
 
 _DATA_URL_RE = re.compile(
-    r"data:(image/[a-z+.-]+);base64,([A-Za-z0-9+/=\s]{64,})")
+    r"data:(image/[a-z+.-]+);base64,([A-Za-z0-9+/=]{64,})")
 
 ARTIFACT_DIR_ENV = "QUORACLE_IMAGE_DIR"
 
