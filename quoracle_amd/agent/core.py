@@ -361,6 +361,16 @@ class AgentActor:
                 session_id=f"{self.state.agent_id}:{model_key}",
             )
             result = await engine.generate(request)
+            if self.runtime.config.trace_prompts:
+                self.runtime.bus.broadcast(
+                    f"agents:{self.state.agent_id}:trace", "llm_exchange", {
+                        "model": model_key, "round": round_num,
+                        "temperature": request.temperature,
+                        "messages": messages,
+                        "response": result.text, "error": result.error,
+                        "input_tokens": result.input_tokens,
+                        "output_tokens": result.output_tokens,
+                        "latency_ms": result.latency_ms})
             if result.error == "context_overflow":
                 # condense once and retry (reference: per_model_query.ex:93-124)
                 await condensation_mod.condense_model_history(

@@ -136,6 +136,7 @@ class LocalEngine(Engine):
                       "decode_tokens": 0, "prefill_tokens": 0,
                       "requests_done": 0, "prefix_hit_tokens": 0}
         self._log_every = int(os.environ.get("QUORACLE_ENGINE_LOG", "0"))
+        self._roctx = bool(os.environ.get("QUORACLE_ROCTX"))
 
     # -- lifecycle -----------------------------------------------------------
 
@@ -516,6 +517,11 @@ class LocalEngine(Engine):
             return torch.tensor(x, dtype=torch.int32, device=dev)
 
         t_fwd = time.monotonic()
+        # rocprof/omnitrace attribution: roctx ranges around each eager
+        # forward; QUORACLE_ROCTX=1 enables
+        if self._roctx:
+            torch.cuda.nvtx.range_push(
+                f"fwd:{hm.key}:T{len(tokens)}:d{n_decode}")
         # batch tensors are H2D copies: build them on the model's stream so
         # the forward (same stream) is ordered after them without a sync
         with hm.stream_ctx():
@@ -533,6 +539,8 @@ class LocalEngine(Engine):
             if sample_rows:
                 rows = torch.tensor(sample_rows, dtype=torch.long, device=dev)
                 logits = hm.model.compute_logits(hidden, rows)
+        if self._roctx:
+            torch.cuda.nvtx.range_pop()
         st = self.stats
         st["engine_steps"] += 1
         st["forward_tokens"] += len(tokens)

@@ -35,6 +35,10 @@ class RuntimeConfig:
     spawn_retries: int = 3
     consensus_retries: int = 3            # ref: message_handler.ex:353-421
     test_mode: bool = False
+    # verbose prompt tracing: broadcast every sent message list + raw
+    # response on agents:<id>:trace (reference: consensus_handler.ex:154-179
+    # debug broadcasts + the show_llm_prompts tooling)
+    trace_prompts: bool = False
 
 
 class TaskRuntime:

@@ -269,3 +269,25 @@ async def test_history_transfer_on_pool_switch():
     root.state.model_histories["fake-x"][0]["content"] = "mutated"
     assert root.state.model_histories["fake-a"][0]["content"] != "mutated"
     await manager.supervisor.terminate_tree(root.state.agent_id)
+
+
+@pytest.mark.asyncio
+async def test_prompt_tracing_broadcasts_full_exchanges():
+    """trace_prompts=True broadcasts every sent message list + raw response
+    (reference: consensus_handler.ex debug broadcasts)."""
+    from quoracle_amd.tasks.runtime import RuntimeConfig
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine,
+                                    config=RuntimeConfig(trace_prompts=True))
+    traces = []
+    result = await manager.create_task("trace me", "default")
+    root = result["root_agent_id"]
+    runtime.bus.on(f"agents:{root}:trace", lambda ev: traces.append(ev))
+    await manager.send_user_message(result["task_id"], "another turn")
+    ok = await wait_until(lambda: len(traces) >= 2)
+    assert ok
+    ev = traces[0]
+    assert ev.payload["model"] in POOL2
+    assert ev.payload["messages"][0]["role"] == "system"
+    assert isinstance(ev.payload["response"], str)
+    await manager.supervisor.terminate_tree(root)
