@@ -37,6 +37,11 @@ def parse_args():
     p.add_argument("--agents-per-gpu", type=int, default=8)
     p.add_argument("--kv-gb", type=float, default=24.0)
     p.add_argument("--device", default=None, help="override (e.g. cpu)")
+    p.add_argument("--scenario", choices=["flat", "tree"], default="flat",
+                   help="flat: independent agents (weak scaling); tree: "
+                        "depth-2 recursive spawn tree per shard "
+                        "(BASELINE config 3)")
+    p.add_argument("--fanout", type=int, default=4)
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: the whole job is ONE "
                         "lockstep TP group hosting a '+'-joined model pool "
@@ -157,15 +162,15 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
         for key in pool_keys:
             pool.assign(key, engine)
 
-    # Build the agent fleet: agents_per_gpu per rank-shard, lockstep-driven
-    # (the actor loop is not started — the bench owns cycle timing).
+    # Build the agent fleet, lockstep-driven (the actor loop is not
+    # started — the bench owns cycle timing).
     actors = []
-    for i in range(args.agents_per_gpu * world):
-        shard = i % world
+
+    def make_agent(name, shard, parent_id, prompt):
         profile = runtime.profiles.resolve(f"bench-r{shard}")
         state = AgentState(
-            agent_id=ids.agent_id(f"bench{i}"),
-            task_id=f"bench-task-{i}", parent_id=None,
+            agent_id=ids.agent_id(name),
+            task_id=f"bench-task-{shard}", parent_id=parent_id,
             profile=f"bench-r{shard}",
             model_pool=list(profile.model_pool),
             capability_groups=[],
@@ -174,12 +179,37 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
         state.init_model_maps()
         actor = AgentActor(state, runtime)
         runtime.registry.register(state.agent_id, actor, state.task_id,
-                                  parent_id=None)
-        actor.state.message_queue.append({
-            "type": "user_message",
-            "content": f"Benchmark task {i}: assess the situation, plan the "
-                       f"work, and coordinate results. Iteration seed {i}."})
+                                  parent_id=parent_id)
+        actor.state.message_queue.append({"type": "user_message",
+                                          "content": prompt})
         actors.append(actor)
+        return actor
+
+    if args.scenario == "tree":
+        # depth-2 recursive tree per shard (BASELINE config 3): the bench
+        # steps EVERY agent of the tree each round; parent<->child messages
+        # flow through the real send_message executor
+        for shard in range(world):
+            root = make_agent(f"root{shard}", shard, None,
+                              f"Coordinate benchmark tree {shard}.")
+            for c in range(args.fanout):
+                child = make_agent(
+                    f"c{shard}-{c}", shard, root.state.agent_id,
+                    f"Subtask {c}: analyze and report to your parent.")
+                root.state.children[child.state.agent_id] = {
+                    "status": "running", "task_description": f"subtask {c}"}
+                for g in range(args.fanout):
+                    gc = make_agent(
+                        f"g{shard}-{c}-{g}", shard, child.state.agent_id,
+                        f"Leaf task {c}.{g}: gather one detail and report.")
+                    child.state.children[gc.state.agent_id] = {
+                        "status": "running",
+                        "task_description": f"leaf {c}.{g}"}
+    else:
+        for i in range(args.agents_per_gpu * world):
+            make_agent(f"bench{i}", i % world, None,
+                       f"Benchmark task {i}: assess the situation, plan the "
+                       f"work, and coordinate results. Iteration seed {i}.")
 
     step_latencies = []
 
@@ -252,6 +282,8 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
             "model": args.model,
             "pool_size": args.pool_size,
             "agents_per_gpu": args.agents_per_gpu,
+            "scenario": args.scenario,
+            "fanout": args.fanout if args.scenario == "tree" else None,
             "agents_total": n_agents,
             "parallelism": parallelism or f"pool-sharded dp{world}",
             "p50_step_latency_ms": round(

@@ -355,6 +355,8 @@ class AgentActor:
                 max_tokens=dynamic_max_tokens(engine, model_key, input_tokens),
                 seed=hash((self.state.agent_id, model_key, round_num)) & 0x7FFFFFFF,
                 action_grammar=True,
+                allowed_actions=self._grammar_actions(),
+                grammar_context=self._grammar_context(),
                 request_id=ids.request_id(),
                 # stable per (agent, model): consecutive cycles reuse the KV
                 # of the unchanged history prefix (engine prefix cache)
@@ -394,6 +396,23 @@ class AgentActor:
                     embed_many=self._embed_many())
             return result.text
         return query_fn
+
+    def _grammar_actions(self) -> List[str]:
+        """Capability-gated action set for constrained decoding; the engine
+        keeps the subset its grammar can template."""
+        from ..governance import profiles as profiles_mod
+        return profiles_mod.allowed_actions(self.state.capability_groups)
+
+    def _grammar_context(self) -> Dict[str, Any]:
+        grove = self.state.grove or {}
+        topology = grove.get("topology") or {}
+        spawn_profile = None
+        for edge in topology.get("edges") or []:
+            inject = edge.get("auto_inject") or {}
+            if isinstance(inject.get("profile"), str):
+                spawn_profile = inject["profile"]
+                break
+        return {"spawn_profile": spawn_profile or self.state.profile}
 
     def _system_prompt(self) -> str:
         if self.state.cached_system_prompt is not None:

@@ -36,6 +36,9 @@ class GenerateRequest:
     # model still does full forwards (the sampler masks logits per template).
     action_grammar: bool = False
     allowed_actions: Optional[List[str]] = None
+    # grammar context: fills parameter literals the model cannot invent
+    # (e.g. the child profile name for spawn_child)
+    grammar_context: Optional[Dict[str, Any]] = None
     request_id: str = ""
     # Prefix-cache key: generate calls sharing a session_id reuse the KV of
     # the longest common token prefix (one session per (agent, model) —

@@ -122,6 +122,9 @@ class LocalEngine(Engine):
             else:
                 per_block = cfg.kv_bytes_per_token() * BLOCK_SIZE
                 blocks = max(8, int(kv_gb_per_model * (1 << 30) / per_block))
+                # small models would otherwise allocate millions of blocks
+                # for the same byte budget; 2M tokens of KV is plenty
+                blocks = min(blocks, (2 << 20) // BLOCK_SIZE)
             self.models[key] = _HostedModel(key, self.device, blocks, tp=tp)
         self.embed_model: Optional[LlamaModel] = None
         if embed_model_key:
@@ -256,8 +259,10 @@ class LocalEngine(Engine):
             max_tokens=max_tokens, seed=request.seed)
         grammar = None
         if request.action_grammar:
-            grammar = ActionGrammar(request.allowed_actions
-                                    or ["orient", "send_message", "todo", "wait"])
+            grammar = ActionGrammar(
+                request.allowed_actions
+                or ["orient", "send_message", "todo", "wait"],
+                context=request.grammar_context)
         gen = None
         if request.seed is not None:
             gen = torch.Generator(device=self.device)

@@ -68,6 +68,7 @@ FORCED, FREE, CHOICE = "forced", "free", "choice"
 # Parameter plans: how to fill each action's required params.
 #   ("free", n)  -> JSON string with n model-sampled tokens
 #   ("lit", x)   -> json literal
+#   ("ctx", key, default) -> JSON string literal from the grammar context
 _PARAM_PLANS: Dict[str, List[Tuple[str, Any]]] = {
     "orient": [("current_situation", ("free", 24)),
                ("goal_clarity", ("free", 12)),
@@ -78,6 +79,12 @@ _PARAM_PLANS: Dict[str, List[Tuple[str, Any]]] = {
                      ("content", ("free", 32))],
     "todo": [("items", ("todo_items", 2))],
     "wait": [("wait", ("lit", True))],
+    "spawn_child": [("task_description", ("free", 24)),
+                    ("success_criteria", ("free", 12)),
+                    ("immediate_context", ("free", 12)),
+                    ("approach_guidance", ("free", 12)),
+                    ("profile", ("ctx", "spawn_profile", "default"))],
+    "file_read": [("path", ("ctx", "file_read_path", "/tmp/notes.txt"))],
 }
 
 
@@ -88,7 +95,10 @@ def _encode(text: str) -> List[int]:
 class ActionGrammar:
     """Per-sequence state machine emitting the next-token constraint."""
 
-    def __init__(self, allowed_actions: Sequence[str], reasoning_tokens: int = 24):
+    def __init__(self, allowed_actions: Sequence[str],
+                 reasoning_tokens: int = 24,
+                 context: Optional[Dict[str, Any]] = None):
+        self.context = context or {}
         self.candidates = [a for a in allowed_actions if a in _PARAM_PLANS] \
             or ["wait"]
         self._ops: List[Tuple] = []
@@ -120,6 +130,9 @@ class ActionGrammar:
                 self._emit_forced('"')
             elif kind == "lit":
                 self._emit_forced(json.dumps(plan[1]))
+            elif kind == "ctx":
+                self._emit_forced(json.dumps(
+                    self.context.get(plan[1], plan[2])))
             elif kind == "todo_items":
                 self._emit_forced('[')
                 for j in range(plan[1]):
