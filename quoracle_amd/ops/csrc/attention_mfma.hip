@@ -427,3 +427,198 @@ paged_attn_prefill_mfma_split_kernel(
     part_acc[base * MF_D + wave * 32 + 16 + c_col] = dead ? 0.f : o_acc1[r];
   }
 }
+
+// ---------------------------------------------------------------------------
+// 32-row Q-tile variant: 8 waves (512 threads) per WG cover 32 query rows,
+// so each staged 64-key K/V chunk serves 2x the query work — half the LDS
+// staging traffic and barrier count per query token vs the 16-row kernel.
+// Wave w: S-phase tile (qblock=w>>2, keyblock=w&3); PV-phase d-columns
+// (w&3)*32..+31 for q rows (w>>2)*16..+15.
+// Grid: (ntiles32, Hq); block 512.  Tile metadata uses 32-row tiles
+// (tile32_* built by the engine alongside the 16-row set).
+// ---------------------------------------------------------------------------
+#define MF2_QT 32
+
+extern "C" __global__ void __launch_bounds__(512)
+paged_attn_prefill_mfma32_kernel(
+    bf16 *__restrict__ out, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+
+  __shared__ bf16 q_s[MF2_QT * KP];
+  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ float s_s[MF2_QT * SP];
+  __shared__ bf16 p_s[MF2_QT * VP];
+  __shared__ float m_s[MF2_QT], l_s[MF2_QT], alpha_s[MF2_QT];
+
+  for (int i = tid; i < MF2_QT * MF_D / 8; i += 512) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
+  if (tid < MF2_QT) {
+    m_s[tid] = -INFINITY;
+    l_s[tid] = 0.f;
+  }
+  __syncthreads();
+
+  f32x4_t o_acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t o_acc1 = {0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int a_row = lane & 15;
+  const int a_koff = (lane >> 4) * 8;
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+  const int qblock = wave >> 2;          // 0..1: 16-row half owned in S & PV
+  const int kblock = wave & 3;           // S-phase key block / PV d-block
+
+  // T14 pipeline at 2 iterations per thread (512 threads)
+  uint4 kreg[2], vreg[2];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+    }
+  };
+
+  issue_loads(0, kv_limit);
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+    write_staged();
+    __syncthreads();
+    if (start + MF_KCHUNK < kv_limit)
+      issue_loads(start + MF_KCHUNK, kv_limit);
+
+    // ---- S = Q·K^T: wave handles (qblock, kblock) ------------------------
+    {
+      f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            q_s + (qblock * 16 + a_row) * KP + kk * 32 + a_koff);
+        bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (kblock * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = qblock * 16 + c_row0 + r;
+        const int col = kblock * 16 + c_col;
+        const int token = start + col;
+        const bool ok = (token <= pos0 + min(row, qn - 1)) && (col < clen);
+        s_s[row * SP + col] = ok ? s_acc[r] * scale : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+    // ---- online softmax (16 threads per row, 32 rows) --------------------
+    {
+      const int row = tid >> 4;
+      const int sub = tid & 15;
+      float v0 = s_s[row * SP + sub];
+      float v1 = s_s[row * SP + sub + 16];
+      float v2 = s_s[row * SP + sub + 32];
+      float v3 = s_s[row * SP + sub + 48];
+      float mymax = fmaxf(fmaxf(v0, v1), fmaxf(v2, v3));
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        mymax = fmaxf(mymax, __shfl_xor(mymax, w, 16));
+      const float m_old = m_s[row];
+      const float mn = fmaxf(m_old, mymax);
+      const float alpha = (m_old == -INFINITY) ? 0.f : __expf(m_old - mn);
+      float p0 = (v0 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v0 - mn);
+      float p1 = (v1 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v1 - mn);
+      float p2 = (v2 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v2 - mn);
+      float p3 = (v3 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v3 - mn);
+      float psum = p0 + p1 + p2 + p3;
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        psum += __shfl_xor(psum, w, 16);
+      if (sub == 0) {
+        l_s[row] = l_s[row] * alpha + psum;
+        m_s[row] = mn;
+        alpha_s[row] = alpha;
+      }
+      p_s[row * VP + sub] = f2bf(p0);
+      p_s[row * VP + sub + 16] = f2bf(p1);
+      p_s[row * VP + sub + 32] = f2bf(p2);
+      p_s[row * VP + sub + 48] = f2bf(p3);
+    }
+    __syncthreads();
+
+    // ---- O rescale + O += P·V: wave covers d cols kblock*32..+31 ---------
+    {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float a = alpha_s[qblock * 16 + c_row0 + r];
+        o_acc0[r] *= a;
+        o_acc1[r] *= a;
+      }
+#pragma unroll
+      for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            p_s + (qblock * 16 + a_row) * VP + kk * 32 + a_koff);
+        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (kblock * 32 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (kblock * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
+        o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = qblock * 16 + c_row0 + r;
+    if (row >= qn) continue;
+    const float denom = l_s[row] > 0.f ? l_s[row] : 1.f;
+    out[((long)(q0 + row) * Hq + h) * MF_D + kblock * 32 + c_col] =
+        f2bf(o_acc0[r] / denom);
+    out[((long)(q0 + row) * Hq + h) * MF_D + kblock * 32 + 16 + c_col] =
+        f2bf(o_acc1[r] / denom);
+  }
+}

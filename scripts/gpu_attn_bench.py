@@ -34,8 +34,20 @@ ms_valu = timeit(lambda: ops.ext().paged_attn_prefill(
 out_valu = out.clone()
 ms_mfma = timeit(lambda: ops.ext().paged_attn_prefill_mfma(
     out, q, kcache, vcache, tables, t0, qn, tseq, tpos, scale))
-rel = (out.float() - out_valu.float()).norm() / out_valu.float().norm()
+out_mfma = out.clone()
+rel = (out_mfma.float() - out_valu.float()).norm() / out_valu.float().norm()
+# 32-row tile metadata for the 8-wave variant
+nt32 = (new + 31) // 32
+t32 = torch.arange(nt32, dtype=torch.int32, device=dev) * 32
+qn32 = torch.clamp(torch.full_like(t32, new) - t32, max=32)
+ts32 = torch.zeros_like(t32)
+tp32 = t32 + cached
+ms_m32 = timeit(lambda: ops.ext().paged_attn_prefill_mfma32(
+    out, q, kcache, vcache, tables, t32, qn32, ts32, tp32, scale))
+rel32 = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
 flops = 2 * 2 * new * (cached + new / 2) * D * Hq
-print(f"VALU prefill: {ms_valu:.3f} ms ({flops/ms_valu/1e9:.1f} TFLOP/s)")
-print(f"MFMA prefill: {ms_mfma:.3f} ms ({flops/ms_mfma/1e9:.1f} TFLOP/s)  "
+print(f"VALU prefill:   {ms_valu:.3f} ms ({flops/ms_valu/1e9:.1f} TFLOP/s)")
+print(f"MFMA16 prefill: {ms_mfma:.3f} ms ({flops/ms_mfma/1e9:.1f} TFLOP/s)  "
       f"speedup {ms_valu/ms_mfma:.1f}x  rel-vs-valu {rel:.4f}")
+print(f"MFMA32 prefill: {ms_m32:.3f} ms ({flops/ms_m32/1e9:.1f} TFLOP/s)  "
+      f"vs16 {ms_mfma/ms_m32:.2f}x  rel-vs-16 {rel32:.4f}")
