@@ -481,6 +481,10 @@ class LocalEngine(Engine):
         tile_qn: List[int] = []
         tile_seq: List[int] = []
         tile_pos0: List[int] = []
+        t32_q0: List[int] = []
+        t32_qn: List[int] = []
+        t32_seq: List[int] = []
+        t32_pos0: List[int] = []
         # decode rows sample at their own row index
         sample_rows: List[int] = list(range(n_decode))
         sample_seqs: List[_Seq] = [s for s in decode if s not in failed]
@@ -503,6 +507,11 @@ class LocalEngine(Engine):
                 tile_qn.append(min(QT, n - off))
                 tile_seq.append(seq_row)
                 tile_pos0.append(cached + off)
+            for off in range(0, n, 2 * QT):
+                t32_q0.append(row + off)
+                t32_qn.append(min(2 * QT, n - off))
+                t32_seq.append(seq_row)
+                t32_pos0.append(cached + off)
             if cached + n == len(seq.known):
                 sample_rows.append(row + n - 1)
                 sample_seqs.append(seq)
@@ -538,7 +547,9 @@ class LocalEngine(Engine):
                 max_ctx=max(ctx_lens) if ctx_lens else 0,
                 max_kv=max_kv,
                 tile_q0=t32(tile_q0), tile_qn=t32(tile_qn),
-                tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0))
+                tile_seq=t32(tile_seq), tile_pos0=t32(tile_pos0),
+                tile32_q0=t32(t32_q0), tile32_qn=t32(t32_qn),
+                tile32_seq=t32(t32_seq), tile32_pos0=t32(t32_pos0))
             hidden = hm.model.forward(batch, hm.kv)
             logits = None
             if sample_rows:

@@ -53,6 +53,11 @@ class ForwardBatch:
     tile_qn: Optional[torch.Tensor] = None
     tile_seq: Optional[torch.Tensor] = None   # row into block_tables
     tile_pos0: Optional[torch.Tensor] = None
+    # 32-row tiling of the same prefill rows (8-wave MFMA kernel)
+    tile32_q0: Optional[torch.Tensor] = None
+    tile32_qn: Optional[torch.Tensor] = None
+    tile32_seq: Optional[torch.Tensor] = None
+    tile32_pos0: Optional[torch.Tensor] = None
     sample_rows: Optional[torch.Tensor] = None  # [R] int64 rows needing logits
 
     @property
@@ -212,7 +217,9 @@ class LlamaModel:
                 self.key, batch.tokens, batch.positions, batch.slots,
                 batch.block_tables, batch.n_decode, batch.ctx_lens,
                 batch.max_ctx, batch.tile_q0, batch.tile_qn, batch.tile_seq,
-                batch.tile_pos0, batch.max_kv, kv.k, kv.v,
+                batch.tile_pos0, batch.tile32_q0, batch.tile32_qn,
+                batch.tile32_seq, batch.tile32_pos0, batch.max_kv,
+                kv.k, kv.v,
                 bool(_os.environ.get("QUORACLE_NO_MFMA_ATTN")))
         cfg = self.lcfg     # per-shard head/intermediate dims under TP
         T = batch.total_tokens

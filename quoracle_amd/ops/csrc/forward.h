@@ -81,7 +81,11 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
                    c10::optional<torch::Tensor> tile_q0,
                    c10::optional<torch::Tensor> tile_qn,
                    c10::optional<torch::Tensor> tile_seq,
-                   c10::optional<torch::Tensor> tile_pos0, int64_t max_kv,
+                   c10::optional<torch::Tensor> tile_pos0,
+                   c10::optional<torch::Tensor> t32_q0,
+                   c10::optional<torch::Tensor> t32_qn,
+                   c10::optional<torch::Tensor> t32_seq,
+                   c10::optional<torch::Tensor> t32_pos0, int64_t max_kv,
                    bool no_mfma) {
   const int64_t D = m.head_dim;
   if (n_decode > 0) {
@@ -119,6 +123,11 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
                                       block_tables, *tile_q0, *tile_qn,
                                       *tile_seq, *tile_pos0, m.scale, pm, pl,
                                       pa);
+      } else if (t32_q0.has_value() && t32_q0->numel() > 0) {
+        // big prefill: 8-wave 32-row tiles (2x K/V reuse per query row)
+        paged_attn_prefill_mfma32(attn_out, q, kcache, vcache, block_tables,
+                                  *t32_q0, *t32_qn, *t32_seq, *t32_pos0,
+                                  m.scale);
       } else {
         paged_attn_prefill_mfma(attn_out, q, kcache, vcache, block_tables,
                                 *tile_q0, *tile_qn, *tile_seq, *tile_pos0,
@@ -171,7 +180,10 @@ inline torch::Tensor forward(
     c10::optional<torch::Tensor> ctx_lens, int64_t max_ctx,
     c10::optional<torch::Tensor> tile_q0, c10::optional<torch::Tensor> tile_qn,
     c10::optional<torch::Tensor> tile_seq,
-    c10::optional<torch::Tensor> tile_pos0, int64_t max_kv,
+    c10::optional<torch::Tensor> tile_pos0,
+    c10::optional<torch::Tensor> t32_q0, c10::optional<torch::Tensor> t32_qn,
+    c10::optional<torch::Tensor> t32_seq,
+    c10::optional<torch::Tensor> t32_pos0, int64_t max_kv,
     std::vector<torch::Tensor> kcaches, std::vector<torch::Tensor> vcaches,
     bool no_mfma) {
   auto it = registry().find(key);
@@ -201,8 +213,8 @@ inline torch::Tensor forward(
     rope_inplace(q, k, positions, m.rope_theta);
     kv_append(kcaches[li], vcaches[li], k, v, slots);
     attend(m, attn_out, q, kcaches[li], vcaches[li], block_tables, n_decode,
-           ctx_lens, max_ctx, tile_q0, tile_qn, tile_seq, tile_pos0, max_kv,
-           no_mfma);
+           ctx_lens, max_ctx, tile_q0, tile_qn, tile_seq, tile_pos0,
+           t32_q0, t32_qn, t32_seq, t32_pos0, max_kv, no_mfma);
     auto proj = at::matmul(attn_out.view({T, q_dim}), L.wo);
     rmsnorm_fused(h, proj, res, L.ffn_norm, m.eps);
     torch::Tensor ffn;
