@@ -32,7 +32,7 @@ from typing import Dict, List, Optional, Sequence, Tuple
 import torch
 
 from ..models import LlamaModel, get_config
-from .api import Engine, GenerateRequest, GenerateResult, MIN_OUTPUT_TOKENS
+from .api import Engine, GenerateRequest, GenerateResult
 from .graphs import DecodeGraphs
 from .kv_cache import BlockManager, OutOfBlocks, Session, SessionCache
 from .sampler import FORCED, ActionGrammar, Sampler, SamplingParams

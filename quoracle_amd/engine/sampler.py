@@ -20,12 +20,12 @@ lib/quoracle/consensus/temperature.ex).
 from __future__ import annotations
 
 import json
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import torch
 
-from .tokenizer import BYTE_VOCAB, EOS
+from .tokenizer import EOS
 
 # bytes allowed in free-text spans inside JSON strings (no '"' or '\\')
 _SAFE_TEXT = [ord(c) for c in

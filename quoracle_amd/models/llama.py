@@ -22,15 +22,14 @@ replacement for that entire layer.
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import torch
 
 from ..ops import dispatch as D
 from .config import ModelConfig, get_config, instance_seed
-from ..parallel.tp import (TPContext, shard_cols, shard_gate_up, shard_qkv,
-                           shard_rows)
+from ..parallel.tp import TPContext, shard_gate_up, shard_qkv, shard_rows
 
 
 @dataclass

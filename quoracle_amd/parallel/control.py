@@ -20,7 +20,7 @@ from __future__ import annotations
 
 import pickle
 import threading
-from typing import Any, Callable, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Sequence
 
 import torch
 import torch.distributed as dist

@@ -17,7 +17,7 @@ overlaps the next prefill by construction.
 
 from __future__ import annotations
 
-from typing import Dict, List, Sequence
+from typing import List, Sequence
 
 import torch
 import torch.distributed as dist
