@@ -1,0 +1,118 @@
+"""External-facing actions: fetch_web (SSRF guard, html ->
+markdown, truncation), call_api (REST auth), call_mcp over a REAL stdio
+JSON-RPC subprocess, answer_engine via the pool."""
+
+import json
+import sys
+import textwrap
+
+import pytest
+
+from quoracle_amd.actions import router as R
+from quoracle_amd.engine.fake import FakeEngine
+
+from helpers import IDLE, action_json, make_runtime
+from test_actions_exec import _actor, _ctx
+
+
+def _http(responses):
+    calls = []
+
+    async def http_fn(method, url, headers=None, data=None):
+        calls.append({"method": method, "url": url, "headers": headers or {},
+                      "data": data})
+        status, body = responses.get(url, (404, "missing"))
+        return status, {}, body
+    http_fn.calls = calls
+    return http_fn
+
+
+@pytest.mark.asyncio
+async def test_fetch_web_html_to_markdown_and_wrapping():
+    runtime = make_runtime()
+    runtime.extras["http_fn"] = _http({
+        "https://example.com/": (200, "<h1>Title</h1><p>Hello <b>web</b></p>")})
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "fetch_web",
+                                      {"url": "https://example.com/"}))
+    assert res["status"] == 200
+    assert "Title" in res["content"] and "<h1>" not in res["content"]
+    assert "NO_EXECUTE_" in json.dumps(res)   # untrusted wrapping
+
+
+@pytest.mark.asyncio
+async def test_fetch_web_ssrf_guard():
+    runtime = make_runtime()
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "fetch_web",
+                                      {"url": "http://169.254.169.254/meta",
+                                       "security_check": True}))
+    assert "ssrf" in json.dumps(res)
+
+
+@pytest.mark.asyncio
+async def test_call_api_bearer_auth_and_json():
+    runtime = make_runtime()
+    runtime.extras["http_fn"] = _http({
+        "https://api.test/v1/thing": (200, json.dumps({"ok": True}))})
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "call_api", {
+        "api_type": "rest", "url": "https://api.test/v1/thing",
+        "method": "GET",
+        "auth": {"auth_type": "bearer", "token": "tok-123"}}))
+    assert json.dumps(res).count("ok")
+    call = runtime.extras["http_fn"].calls[0]
+    assert call["headers"].get("Authorization") == "Bearer tok-123"
+
+
+@pytest.mark.asyncio
+async def test_call_mcp_stdio_roundtrip(tmp_path):
+    """Full MCP lifecycle against a real stdio JSON-RPC subprocess."""
+    server = tmp_path / "mcp_echo.py"
+    server.write_text(textwrap.dedent("""
+        import json, sys
+        for line in sys.stdin:
+            req = json.loads(line)
+            rid = req.get("id")
+            if rid is None:
+                continue
+            method = req.get("method")
+            if method == "initialize":
+                result = {"serverInfo": {"name": "echo"}}
+            elif method == "tools/call":
+                p = req.get("params", {})
+                result = {"content": [{"type": "text",
+                                       "text": "echo:" +
+                                       json.dumps(p.get("arguments"))}]}
+            else:
+                result = {}
+            sys.stdout.write(json.dumps({"jsonrpc": "2.0", "id": rid,
+                                         "result": result}) + "\\n")
+            sys.stdout.flush()
+    """))
+    runtime = make_runtime()
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+        "transport": "stdio",
+        "command": f"{sys.executable} {server}"}))
+    cid = res.get("connection_id")
+    assert cid, res
+    res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+        "connection_id": cid, "tool": "shout",
+        "arguments": {"msg": "hi"}}))
+    assert "echo:" in json.dumps(res)
+    res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+        "connection_id": cid, "terminate": True}))
+    assert res["status"] == "terminated"
+
+
+@pytest.mark.asyncio
+async def test_answer_engine_uses_configured_role_model():
+    runtime = make_runtime(engine=FakeEngine(
+        default_response="Paris is the capital of France."))
+    runtime.config.model_roles["answer_engine"] = "fake-b"
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "answer_engine",
+                                      {"prompt": "capital of France?"}))
+    assert "Paris" in json.dumps(res)
+    assert res.get("model") == "fake-b" or "fake-b" in json.dumps(res)
