@@ -26,6 +26,7 @@ class CreateTaskBody(BaseModel):
     budget_limit: Optional[float] = None
     global_context: Optional[str] = None
     role: Optional[str] = None
+    grove: Optional[str] = None      # grove name under the groves dir
 
 
 class MessageBody(BaseModel):
@@ -65,12 +66,27 @@ def create_app(manager) -> FastAPI:
     @app.post("/api/tasks")
     async def create_task(body: CreateTaskBody):
         from ..tasks.manager import TaskError
+        grove = None
+        if body.grove:
+            from ..governance import groves as groves_mod
+            import os as _os
+            base = runtime.config.groves_dir or "groves"
+            try:
+                grove = groves_mod.load_grove(_os.path.join(base, body.grove))
+            except (OSError, ValueError) as exc:
+                raise HTTPException(400, f"bad_grove: {exc}")
         try:
             return await manager.create_task(
                 body.prompt, body.profile, budget_limit=body.budget_limit,
-                global_context=body.global_context, role=body.role)
+                global_context=body.global_context, role=body.role,
+                grove=grove)
         except TaskError as exc:
             raise HTTPException(400, exc.reason)
+
+    @app.get("/api/groves")
+    def groves():
+        from ..governance import groves as groves_mod
+        return groves_mod.list_groves(runtime.config.groves_dir or "groves")
 
     @app.get("/api/tasks/{task_id}/tree")
     def task_tree(task_id: str):

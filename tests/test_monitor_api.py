@@ -62,3 +62,20 @@ def test_dashboard_served(client):
     c, _ = client
     r = c.get("/")
     assert r.status_code == 200 and "quoracle" in r.text
+
+
+def test_grove_task_creation(client, monkeypatch):
+    c, runtime = client
+    import os
+    runtime.config.groves_dir = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "groves")
+    names = {g["name"] for g in c.get("/api/groves").json()}
+    assert "qa-benchmark" in names
+    r = c.post("/api/tasks", json={"prompt": "run it", "profile": "default",
+                                   "grove": "qa-benchmark"})
+    assert r.status_code == 200
+    root = r.json()["root_agent_id"]
+    actor = runtime.registry.lookup(root).actor
+    assert (actor.state.grove or {}).get("name") == "qa-benchmark"
+    r = c.post("/api/tasks", json={"prompt": "x", "grove": "no-such"})
+    assert r.status_code == 400
