@@ -34,21 +34,33 @@ def merge_lessons(
     if not new:
         return merged[:max_lessons]
 
+    sim = None
+    existing_vecs = new_vecs = None
     if embed_many is not None and merged:
         texts = [l.get("text", "") for l in merged] + \
                 [l.get("text", "") for l in new]
-        vectors = embed_many(texts)
-        existing_vecs = vectors[: len(merged)]
-        new_vecs = vectors[len(merged):]
-    else:
-        existing_vecs = new_vecs = None
+        if getattr(embed_many, "has_similarity", False):
+            # one embedding forward + fused cosine kernel for ALL pairs
+            sim = embed_many.similarity_matrix(texts)
+        else:
+            vectors = embed_many(texts)
+            existing_vecs = vectors[: len(merged)]
+            new_vecs = vectors[len(merged):]
 
+    n_existing = len(merged)
+    # rows[j] = this merged entry's row in the similarity matrix (sim path);
+    # front-inserts during the loop keep both structures aligned
+    rows = list(range(n_existing))
     for i, lesson in enumerate(new):
         duplicate_idx = None
         for j, old in enumerate(merged):
-            if existing_vecs is not None:
-                sim = cosine_similarity(new_vecs[i], existing_vecs[j])
-                if sim >= threshold:
+            if sim is not None:
+                if sim[n_existing + i][rows[j]] >= threshold:
+                    duplicate_idx = j
+                    break
+            elif existing_vecs is not None:
+                s_ij = cosine_similarity(new_vecs[i], existing_vecs[j])
+                if s_ij >= threshold:
                     duplicate_idx = j
                     break
             elif old.get("text") == lesson.get("text"):
@@ -59,6 +71,7 @@ def merge_lessons(
                 merged[duplicate_idx].get("confidence", 1) + 1
         else:
             merged.insert(0, dict(lesson))
+            rows.insert(0, n_existing + i)
             if existing_vecs is not None:
                 existing_vecs = [new_vecs[i]] + list(existing_vecs)
 

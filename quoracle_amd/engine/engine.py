@@ -221,6 +221,25 @@ class LocalEngine(Engine):
         norm = hid.norm(dim=-1, keepdim=True).clamp_min(1e-8)
         return (hid / norm).cpu().tolist()
 
+    def similarity_matrix(self, texts: List[str]):
+        """Pairwise cosine matrix of the texts' embeddings — embedding
+        forward + the fused cosine_sim_matrix kernel in one GPU pass
+        (the consensus vote's hot path)."""
+        if self.embed_model is None:
+            raise RuntimeError("no embedding model hosted")
+        batches = []
+        for t in texts:
+            ids_ = self.tokenizer.encode(t)[:512] or [EOS]
+            batches.append(torch.tensor(ids_, dtype=torch.int32,
+                                        device=self.device))
+        from ..ops import dispatch as D
+        with self._gpu_lock:
+            hid = self.embed_model.embed_texts_hidden(batches).float()
+            out = torch.empty((hid.shape[0], hid.shape[0]),
+                              dtype=torch.float32, device=hid.device)
+            D.cosine_sim_matrix(out, hid)
+        return out.cpu().tolist()
+
     def count_tokens(self, text: str) -> int:
         return self.tokenizer.count(text)
 

@@ -118,6 +118,14 @@ class FakeEngine:
         self.embed_calls.append(list(texts))
         return [deterministic_embedding(t) for t in texts]
 
+    def similarity_matrix(self, texts: List[str]):
+        vecs = [deterministic_embedding(t) for t in texts]
+        self.embed_calls.append(list(texts))
+        out = []
+        for a in vecs:
+            out.append([sum(x * y for x, y in zip(a, b)) for b in vecs])
+        return out
+
     def count_tokens(self, text: str) -> int:
         return max(1, len(text) // 4) if text else 0
 

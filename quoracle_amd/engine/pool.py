@@ -43,7 +43,36 @@ class EnginePool:
         return self._embedder
 
     def embed_many_sync(self, texts: List[str]) -> Sequence[Sequence[float]]:
-        return sync_embed_many(self.embedder)(texts)
+        return self.embed_facade(texts)
+
+    @property
+    def embed_facade(self) -> "EmbedFacade":
+        return EmbedFacade(self.embedder)
 
     def models(self) -> List[str]:
         return list(self._by_model)
+
+
+class EmbedFacade:
+    """Callable embed_many with an optional fused similarity_matrix —
+    components that only batch-compare texts (lesson dedup, clustering)
+    use ONE embedding forward + cosine kernel pass instead of embedding
+    then pairwise python cosine."""
+
+    def __init__(self, engine):
+        self._engine = engine
+        self._embed = sync_embed_many(engine)
+
+    def __call__(self, texts: List[str]):
+        return self._embed(list(texts))
+
+    @property
+    def has_similarity(self) -> bool:
+        return hasattr(self._engine, "similarity_matrix")
+
+    def similarity_matrix(self, texts: List[str]):
+        if self.has_similarity:
+            return self._engine.similarity_matrix(list(texts))
+        from ..consensus.rules import cosine_similarity
+        vecs = self(list(texts))
+        return [[cosine_similarity(a, b) for b in vecs] for a in vecs]

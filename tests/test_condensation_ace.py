@@ -101,3 +101,22 @@ def test_lesson_dedup_cosine_and_prune():
              "confidence": 1} for i in range(130)]
     pruned = merge_lessons([], many, embed)
     assert len(pruned) <= 100
+
+
+def test_lesson_dedup_via_fused_similarity_facade():
+    """With an engine exposing similarity_matrix (the GPU cosine-kernel
+    seam), dedup uses one batched pass — including new-vs-new pairs."""
+    from quoracle_amd.engine.fake import FakeEngine
+    from quoracle_amd.engine.pool import EnginePool
+    pool = EnginePool(default=FakeEngine(), embedder=FakeEngine())
+    facade = pool.embed_facade
+    assert facade.has_similarity
+    base = [{"text": "check the logs before restarting", "confidence": 1}]
+    new = [{"text": "check the logs before restarting", "confidence": 1},
+           {"text": "a totally different insight about caching", "confidence": 1},
+           {"text": "a totally different insight about caching", "confidence": 1}]
+    merged = merge_lessons(base, new, facade)
+    texts = [l["text"] for l in merged]
+    # exact duplicate merged into base; the two identical new ones merged
+    assert len(merged) == 2
+    assert merged[0]["confidence"] >= 2 or merged[1]["confidence"] >= 2
