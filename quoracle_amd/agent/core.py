@@ -293,7 +293,9 @@ class AgentActor:
 
     def _embed_many(self):
         try:
-            return self.runtime.engines.embed_many_sync
+            # the facade is callable (embed_many) AND carries the fused
+            # similarity_matrix used by lesson dedup / clustering
+            return self.runtime.engines.embed_facade
         except RuntimeError:
             return None
 
