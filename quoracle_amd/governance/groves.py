@@ -341,15 +341,28 @@ def substitute_grove_vars(value: Any, grove_vars: Dict[str, str]) -> Any:
 # ---------------------------------------------------------------------------
 
 def select_schema(schemas: Optional[List[Dict[str, Any]]], file_path: str,
-                  workspace: Optional[str]) -> Optional[Dict[str, Any]]:
-    """Most-specific matching path_pattern wins."""
-    if not schemas or not workspace:
+                  workspace: Optional[str],
+                  grove_path: Optional[str] = None) -> Optional[Dict[str, Any]]:
+    """Most-specific matching path_pattern wins.  Patterns are relative to
+    the workspace when the file lives there, else to the grove root."""
+    if not schemas:
         return None
     expanded = os.path.abspath(file_path)
-    ws = os.path.abspath(workspace).rstrip("/") + "/"
-    if not expanded.startswith(ws):
+    bases = []
+    if workspace:
+        ws = workspace if os.path.isabs(workspace) else \
+            os.path.join(grove_path or ".", workspace)
+        bases.append(os.path.abspath(ws))
+    if grove_path:
+        bases.append(os.path.abspath(grove_path))
+    relative = None
+    for base in bases:
+        prefix = base.rstrip("/") + "/"
+        if expanded.startswith(prefix):
+            relative = expanded[len(prefix):]
+            break
+    if relative is None:
         return None
-    relative = expanded[len(ws):]
     matching = [s for s in schemas
                 if isinstance(s.get("path_pattern"), str)
                 and path_matches_pattern(relative, s["path_pattern"])]
@@ -415,10 +428,10 @@ def validate_file_write(grove: Optional[Dict[str, Any]], file_path: str,
     if not grove:
         return
     schema_entry = select_schema(grove.get("schemas"), file_path,
-                                 grove.get("workspace"))
+                                 grove.get("workspace"), grove.get("path"))
     if schema_entry is None:
         return
-    definition = schema_entry.get("definition")
+    definition = schema_entry.get("definition") or schema_entry.get("schema")
     if isinstance(definition, str):
         schema_json = safe_read_file(definition, grove.get("path", "."))
         schema = json.loads(schema_json)

@@ -182,6 +182,35 @@ def create_app(manager) -> FastAPI:
         runtime.vault.put(body.name, body.value, body.description)
         return {"ok": True}
 
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus-format counters (the reference's Telemetry /
+        LiveDashboard metrics layer, reference: quoracle_web/telemetry.ex)."""
+        from fastapi.responses import PlainTextResponse
+        lines = []
+
+        def emit(name, value, help_=""):
+            if help_:
+                lines.append(f"# HELP {name} {help_}")
+            lines.append(f"# TYPE {name} gauge")
+            lines.append(f"{name} {value}")
+
+        emit("quoracle_agents_alive", len(runtime.registry.all_ids()),
+             "live agents in the registry")
+        tasks_ = runtime.store.list_tasks()
+        emit("quoracle_tasks_total", len(tasks_))
+        emit("quoracle_tasks_running",
+             sum(1 for t in tasks_ if t.get("status") == "running"))
+        embedder = getattr(runtime.engines, "_embedder", None)
+        for key, val in (getattr(embedder, "stats", None) or {}).items():
+            emit(f"quoracle_engine_{key}", val)
+        total = runtime.store.total_cost(
+            [a for t in tasks_ for a in
+             [r["agent_id"] for r in
+              runtime.store.agents_for_task(t["task_id"])]])
+        emit("quoracle_cost_usd_total", round(total, 6))
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/api/engine/stats")
     def engine_stats():
         stats = {}

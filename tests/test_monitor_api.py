@@ -79,3 +79,12 @@ def test_grove_task_creation(client, monkeypatch):
     assert (actor.state.grove or {}).get("name") == "qa-benchmark"
     r = c.post("/api/tasks", json={"prompt": "x", "grove": "no-such"})
     assert r.status_code == 400
+
+
+def test_prometheus_metrics(client):
+    c, runtime = client
+    c.post("/api/tasks", json={"prompt": "m", "profile": "default"})
+    body = c.get("/metrics").text
+    assert "quoracle_agents_alive" in body
+    assert "quoracle_tasks_running 1" in body
+    assert "quoracle_cost_usd_total" in body
