@@ -37,6 +37,12 @@ def parse_args():
     p.add_argument("--agents-per-gpu", type=int, default=8)
     p.add_argument("--kv-gb", type=float, default=24.0)
     p.add_argument("--device", default=None, help="override (e.g. cpu)")
+    p.add_argument("--pool-scope", choices=["shard", "global"],
+                   default="shard",
+                   help="shard: each agent votes over its own rank's models "
+                        "(weak scaling). global: ONE pool spanning every "
+                        "rank, one model per GPU (BASELINE config 4; use "
+                        "--model a+b+... for a heterogeneous pool)")
     p.add_argument("--scenario", choices=["flat", "tree"], default="flat",
                    help="flat: independent agents (weak scaling); tree: "
                         "depth-2 recursive spawn tree per shard "
@@ -51,7 +57,11 @@ def parse_args():
 
 
 def rank_model_keys(model: str, pool_size: int, rank: int):
-    return [f"{model}#r{rank}m{j}" for j in range(pool_size)]
+    """Shard mode: rank r hosts pool_size instances of the preset.
+    '+'-joined presets round-robin over the instances (heterogeneous)."""
+    presets = model.split("+")
+    return [f"{presets[j % len(presets)]}#r{rank}m{j}"
+            for j in range(pool_size)]
 
 
 T_START = time.perf_counter()
@@ -103,7 +113,12 @@ def main():
         print(json.dumps(result), flush=True)
         return
 
-    local_keys = rank_model_keys(args.model, args.pool_size, rank)
+    if args.pool_scope == "global":
+        # config 4: one model per GPU, all of them in ONE consensus pool
+        presets = args.model.split("+")
+        local_keys = [f"{presets[rank % len(presets)]}#r{rank}"]
+    else:
+        local_keys = rank_model_keys(args.model, args.pool_size, rank)
     engine = LocalEngine(
         local_keys, device=device, kv_gb_per_model=args.kv_gb,
         embed_model_key="embed-small")   # every rank joins the RCCL vote
@@ -142,8 +157,15 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
         # RCCL all-gather over xGMI
         pool = EnginePool(embedder=DistributedEmbedder(engine, client, world))
         pool._embedder_engine = engine
+    presets = args.model.split("+")
+
+    def keys_for_rank(r):
+        if args.pool_scope == "global":
+            return [f"{presets[r % len(presets)]}#r{r}"]
+        return rank_model_keys(args.model, args.pool_size, r)
+
     for r in range(world):
-        for key in rank_model_keys(args.model, args.pool_size, r):
+        for key in keys_for_rank(r):
             if r == 0:
                 pool.assign(key, engine)
             else:
@@ -152,11 +174,13 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
 
     runtime = TaskRuntime(engines=pool, config=RuntimeConfig())
     Supervisor(runtime)
+    global_pool = [k for r in range(world) for k in keys_for_rank(r)]
     for r in range(world):
         runtime.profiles.put(Profile(
             name=f"bench-r{r}", description="bench pool shard",
-            model_pool=pool_keys or rank_model_keys(args.model,
-                                                    args.pool_size, r),
+            model_pool=pool_keys or (
+                global_pool if args.pool_scope == "global"
+                else keys_for_rank(r)),
             capability_groups=[], max_refinement_rounds=2))
     if pool_keys:
         for key in pool_keys:
@@ -283,6 +307,7 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
             "pool_size": args.pool_size,
             "agents_per_gpu": args.agents_per_gpu,
             "scenario": args.scenario,
+            "pool_scope": args.pool_scope,
             "fanout": args.fanout if args.scenario == "tree" else None,
             "agents_total": n_agents,
             "parallelism": parallelism or f"pool-sharded dp{world}",
