@@ -149,28 +149,38 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
 
 inline torch::Tensor moe_ffn(const Model &m, const Layer &L,
                              torch::Tensor h) {
+  // token-sorted dispatch (mirrors models/llama.py _moe_ffn): one
+  // contiguous GEMM slice per expert, single weighted scatter-add
   auto logits = at::matmul(h, L.router).to(torch::kFloat32);
   auto probs = at::softmax(logits, -1);
   auto topk = probs.topk(m.top_k, -1);
   auto weights = std::get<0>(topk);
   auto experts = std::get<1>(topk);
   weights = (weights / weights.sum(-1, true)).to(h.scalar_type());
-  auto out = torch::zeros_like(h);
   const int64_t E = L.router.size(1);
+  auto expert_flat = experts.reshape({-1});
+  auto order = expert_flat.argsort(/*stable=*/true);
+  auto token_of = order.div(m.top_k, "floor");
+  auto h_sorted = h.index_select(0, token_of);
+  auto counts =
+      at::bincount(expert_flat, /*weights=*/{}, /*minlength=*/E).cpu();
+  auto counts_a = counts.accessor<int64_t, 1>();
+  auto down_sorted = torch::empty_like(h_sorted);
+  int64_t start = 0;
   for (int64_t e = 0; e < E; ++e) {
-    auto mask = experts.eq(e);
-    auto rows = mask.any(-1).nonzero().squeeze(-1);
-    if (rows.numel() == 0) continue;
-    auto he = h.index_select(0, rows);
+    const int64_t n = counts_a[e];
+    if (n == 0) continue;
+    auto he = h_sorted.narrow(0, start, n);
     auto gu = at::matmul(he, L.w_gate_up[e]);
-    auto act = torch::empty({he.size(0), m.intermediate},
-                            h.options());
+    auto act = torch::empty({n, m.intermediate}, h.options());
     swiglu(act, gu);
-    auto down = at::matmul(act, L.w_down[e]);
-    auto w_e = (weights * mask.to(weights.scalar_type()))
-                   .sum(-1).index_select(0, rows);
-    out.index_add_(0, rows, down * w_e.unsqueeze(1));
+    down_sorted.narrow(0, start, n) = at::matmul(act, L.w_down[e]);
+    start += n;
   }
+  auto w_sorted =
+      weights.reshape({-1}).index_select(0, order).unsqueeze(1);
+  auto out = torch::zeros_like(h);
+  out.index_add_(0, token_of, down_sorted * w_sorted);
   return out;
 }
 

@@ -116,3 +116,20 @@ def test_cpp_forward_matches_python(dev):
     l_cpp = m_cpp.compute_logits(h_cpp, rows)
     l_py = m_py.compute_logits(h_py, rows)
     assert torch.allclose(l_cpp, l_py, atol=1e-2, rtol=1e-2)
+
+
+def test_moe_generate_on_gpu(dev):
+    """tiny-moe through the engine on GPU: exercises the C++ sorted-MoE
+    dispatch end-to-end (router -> expert GEMM slices -> scatter-add)."""
+    from quoracle_amd.engine.api import GenerateRequest
+    from quoracle_amd.engine.engine import LocalEngine
+    engine = LocalEngine(["tiny-moe#g"], device=dev, kv_blocks_override=512,
+                         embed_model_key=None)
+    r = engine.generate_sync(GenerateRequest(
+        model_key="tiny-moe#g",
+        messages=[{"role": "user", "content": "hello"}],
+        temperature=0.7, max_tokens=300, seed=9,
+        action_grammar=True, session_id="moe-g"), timeout=300)
+    assert r.ok, r.error
+    parsed = json.loads(r.text)
+    assert parsed["action"] in {"orient", "send_message", "todo", "wait"}

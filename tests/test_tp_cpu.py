@@ -82,3 +82,20 @@ def test_tp2_matches_single(moe):
     assert all(p.exitcode == 0 for p in procs)
     rel = (tp_hidden - single).norm() / single.norm()
     assert rel < 0.05, f"TP mismatch rel={rel}"
+
+
+def test_moe_sorted_dispatch_matches_naive():
+    """Token-sorted MoE == per-expert masked loop (CPU, fp32-ish check)."""
+    import torch
+    from dataclasses import replace
+    from quoracle_amd.models import LlamaModel
+    from quoracle_amd.models.config import PRESETS
+    cfg = replace(PRESETS["tiny"], n_experts=4, top_k_experts=2)
+    model = LlamaModel("moe-sort", torch.device("cpu"), cfg=cfg)
+    torch.manual_seed(3)
+    h = torch.randn(37, cfg.hidden, dtype=torch.bfloat16)
+    layer = model.layers[0]
+    a = model._moe_ffn(h, layer).float()
+    b = model._moe_ffn_naive(h, layer).float()
+    assert torch.allclose(a, b, atol=2e-2, rtol=2e-2), \
+        (a - b).abs().max().item()
