@@ -545,6 +545,17 @@ class AgentActor:
                 conn.proc.terminate()
             except Exception:
                 pass
+        # release the engine-side prefix-cache sessions (KV blocks) this
+        # agent held; histories are persisted, so a future restore simply
+        # re-prefills through the prefix cache
+        for model in self.state.model_pool:
+            try:
+                engine = self.runtime.engines.engine_for(model)
+            except KeyError:
+                continue
+            drop = getattr(engine, "drop_session", None)
+            if drop is not None:
+                drop(model, f"{self.state.agent_id}:{model}")
         self._persist()
         self.runtime.store.update_agent_status(self.state.agent_id, "terminated")
         self.runtime.registry.unregister(self.state.agent_id)

@@ -176,3 +176,46 @@ def test_moe_model_generates_through_engine():
     assert r.ok, r.error
     parsed = json.loads(r.text)
     assert parsed["action"] in {"orient", "send_message", "todo", "wait"}
+
+
+def test_agent_termination_frees_sessions():
+    """Terminating an agent releases its engine sessions' KV blocks
+    (histories persist; restore re-prefills via the prefix cache)."""
+    import asyncio
+    from helpers import make_manager, IDLE
+    eng = LocalEngine(["tiny#t0", "tiny#t1"], device=torch.device("cpu"),
+                      kv_blocks_override=2048, embed_model_key=None,
+                      prefill_chunk=64)
+
+    async def run():
+        from quoracle_amd.engine.pool import EnginePool
+        from quoracle_amd.tasks.manager import TaskManager
+        from quoracle_amd.tasks.runtime import TaskRuntime
+        from quoracle_amd.agent.supervisor import Supervisor
+        from quoracle_amd.governance.profiles import Profile
+        eng.start()
+        try:
+            pool = EnginePool(embedder=eng)
+            pool.assign("tiny#t0", eng)
+            pool.assign("tiny#t1", eng)
+            runtime = TaskRuntime(engines=pool)
+            Supervisor(runtime)
+            runtime.profiles.put(Profile(
+                name="p", description="", model_pool=["tiny#t0", "tiny#t1"],
+                capability_groups=[]))
+            manager = TaskManager(runtime)
+            result = await manager.create_task("session hygiene", "p")
+            root = result["root_agent_id"]
+            import time
+            deadline = time.monotonic() + 60
+            actor = runtime.registry.lookup(root).actor
+            while actor.steps_completed < 1 and time.monotonic() < deadline:
+                await asyncio.sleep(0.05)
+            assert actor.steps_completed >= 1
+            free_before = eng.models["tiny#t0"].mgr.free_blocks
+            await manager.supervisor.terminate_tree(root)
+            free_after = eng.models["tiny#t0"].mgr.free_blocks
+            assert free_after > free_before
+        finally:
+            eng.stop()
+    asyncio.run(run())
