@@ -44,7 +44,7 @@ def parse_response(text: str) -> Dict[str, Any]:
 
     wait = parsed.get("wait")
     if not (wait is None or isinstance(wait, bool) or
-            (isinstance(wait, int) and not isinstance(wait, bool))):
+            (isinstance(wait, (int, float)) and not isinstance(wait, bool))):
         wait = None
 
     result: Dict[str, Any] = {
