@@ -167,7 +167,8 @@ class Supervisor:
             if actor is None:
                 raise SpawnError(f"start_failed: {last_exc}")
 
-            initial = _initial_message(params)
+            narrative = await self._ancestor_narrative(parent)
+            initial = _initial_message(params, narrative)
             await actor.deliver({"type": "user_message", "content": initial})
             state.children[child_id]["status"] = "running"
             await parent.deliver({"type": "child_spawned", "child_id": child_id})
@@ -180,6 +181,51 @@ class Supervisor:
             state.children.pop(child_id, None)
             await parent.deliver({"type": "spawn_failed", "child_id": child_id,
                                   "reason": str(exc)})
+
+    async def _ancestor_narrative(self, parent: AgentActor) -> str:
+        """Lineage context for the child: the parent's recent decision
+        trail, LLM-summarized through the summarization-role model when it
+        is long (reference: spawn/config_builder.ex:125,230); truncation is
+        the fallback when no summarizer is reachable."""
+        state = parent.state
+        history = max(state.model_histories.values(), key=len, default=[])
+        decisions = [e for e in history if e.get("type") == "decision"][:6]
+        own_task = ""
+        for e in reversed(history):              # oldest entries last in scan
+            if e.get("type") in ("prompt", "event"):
+                own_task = str(e.get("content", ""))[:300]
+                break
+        if not decisions and not own_task:
+            return ""
+        lines = [f"Parent's task: {own_task}" if own_task else ""]
+        for e in reversed(decisions):            # oldest first
+            c = e.get("content")
+            if isinstance(c, dict):
+                lines.append(f"- {c.get('action')}: "
+                             f"{str(c.get('reasoning', ''))[:200]}")
+        raw = "\n".join(l for l in lines if l)
+        if len(raw) <= 1200:
+            return raw
+        model = (self.runtime.config.model_roles.get("summarization")
+                 or state.model_pool[0] if state.model_pool else None)
+        if model is None:
+            return raw[:1200]
+        try:
+            from ..engine.api import GenerateRequest
+            engine = self.runtime.engines.engine_for(model)
+            result = await engine.generate(GenerateRequest(
+                model_key=model,
+                messages=[{"role": "user",
+                           "content": "Summarize this agent's work so far "
+                                      "in <=5 bullet points for a child "
+                                      "agent inheriting a subtask:\n"
+                                      + raw[:6000]}],
+                temperature=0.3, max_tokens=512))
+            if result.ok and result.text.strip():
+                return result.text.strip()
+        except Exception:  # noqa: BLE001 — narrative is best-effort
+            pass
+        return raw[:1200]
 
     def _build_child_config(self, parent: AgentActor, child_id: str, profile,
                             params: Dict[str, Any], inject: Dict[str, Any],
@@ -282,8 +328,10 @@ class Supervisor:
         return len(window) <= MAX_RESTARTS
 
 
-def _initial_message(params: Dict[str, Any]) -> str:
+def _initial_message(params: Dict[str, Any], narrative: str = "") -> str:
     parts = [f"# Task\n{params.get('task_description', '')}"]
+    if narrative:
+        parts.append(f"# Lineage context\n{narrative}")
     if params.get("success_criteria"):
         parts.append(f"# Success criteria\n{params['success_criteria']}")
     if params.get("immediate_context"):

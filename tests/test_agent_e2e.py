@@ -291,3 +291,39 @@ async def test_prompt_tracing_broadcasts_full_exchanges():
     assert ev.payload["messages"][0]["role"] == "system"
     assert isinstance(ev.payload["response"], str)
     await manager.supervisor.terminate_tree(root)
+
+
+@pytest.mark.asyncio
+async def test_child_initial_message_carries_lineage_context():
+    """Spawned children receive the parent's summarized decision trail
+    (reference: spawn/config_builder.ex ancestor narrative)."""
+    engine = FakeEngine(default_response=IDLE)
+    spawn = action_json("spawn_child", {
+        "task_description": "sub work", "success_criteria": "done",
+        "immediate_context": "ctx", "approach_guidance": "go",
+        "profile": "default"})
+    for m in POOL2:
+        engine.push_response(m, action_json(
+            "orient", {"current_situation": "investigating the outage",
+                       "goal_clarity": "clear",
+                       "available_resources": "logs",
+                       "key_challenges": "none",
+                       "delegation_consideration": "spawn soon"}))
+    for m in POOL2:
+        engine.push_response(m, spawn)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("lineage root task", "default")
+    root_id = result["root_agent_id"]
+    assert await wait_until(
+        lambda: runtime.registry.children_of(root_id), timeout=10)
+    child_id = runtime.registry.children_of(root_id)[0]
+    child = runtime.registry.lookup(child_id).actor
+
+    def child_got_initial():
+        h = child.state.model_histories.get(POOL2[0], [])
+        return any("Lineage context" in str(e.get("content", "")) for e in h)
+    assert await wait_until(child_got_initial, timeout=10)
+    h = child.state.model_histories[POOL2[0]]
+    joined = "\n".join(str(e.get("content")) for e in h)
+    assert "orient" in joined        # the parent's decision trail came along
+    await manager.supervisor.terminate_tree(root_id)
