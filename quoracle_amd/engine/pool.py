@@ -12,8 +12,49 @@ from __future__ import annotations
 
 from typing import Any, Dict, List, Optional, Sequence
 
+import hashlib
+import time as _time
+
 from .api import Engine
 from .fake import sync_embed_many
+
+# reference parity: embeddings.ex:24-25,402-426 — SHA-keyed cache,
+# 1h TTL, 1000 entries, LRU eviction of the oldest 10% under pressure
+EMBED_CACHE_TTL_S = 3600.0
+EMBED_CACHE_MAX = 1000
+
+
+class EmbedCache:
+    def __init__(self, ttl_s: float = EMBED_CACHE_TTL_S,
+                 max_entries: int = EMBED_CACHE_MAX):
+        self.ttl_s = ttl_s
+        self.max_entries = max_entries
+        self._data = {}          # sha -> (ts, vector)
+        self.hits = 0
+        self.misses = 0
+
+    @staticmethod
+    def key(text: str) -> str:
+        return hashlib.sha256(text.encode("utf-8", "replace")).hexdigest()
+
+    def get(self, text: str):
+        k = self.key(text)
+        entry = self._data.get(k)
+        if entry is None or _time.monotonic() - entry[0] > self.ttl_s:
+            self.misses += 1
+            self._data.pop(k, None)
+            return None
+        self.hits += 1
+        # LRU touch
+        self._data[k] = (_time.monotonic(), entry[1])
+        return entry[1]
+
+    def put(self, text: str, vector) -> None:
+        if len(self._data) >= self.max_entries:
+            evict = max(1, self.max_entries // 10)
+            for k in sorted(self._data, key=lambda k: self._data[k][0])[:evict]:
+                self._data.pop(k, None)
+        self._data[self.key(text)] = (_time.monotonic(), vector)
 
 
 class EnginePool:
@@ -26,6 +67,7 @@ class EnginePool:
         self._default = default
         self._by_model = by_model or {}
         self._embedder = embedder or default or next(iter(self._by_model.values()), None)
+        self._embed_cache = EmbedCache()
 
     def engine_for(self, model_key: str) -> Engine:
         engine = self._by_model.get(model_key, self._default)
@@ -47,7 +89,7 @@ class EnginePool:
 
     @property
     def embed_facade(self) -> "EmbedFacade":
-        return EmbedFacade(self.embedder)
+        return EmbedFacade(self.embedder, self._embed_cache)
 
     def models(self) -> List[str]:
         return list(self._by_model)
@@ -59,12 +101,29 @@ class EmbedFacade:
     use ONE embedding forward + cosine kernel pass instead of embedding
     then pairwise python cosine."""
 
-    def __init__(self, engine):
+    def __init__(self, engine, cache: Optional[EmbedCache] = None):
         self._engine = engine
         self._embed = sync_embed_many(engine)
+        self._cache = cache
 
     def __call__(self, texts: List[str]):
-        return self._embed(list(texts))
+        texts = list(texts)
+        if self._cache is None:
+            return self._embed(texts)
+        out = [None] * len(texts)
+        missing = []
+        for i, t in enumerate(texts):
+            vec = self._cache.get(t)
+            if vec is None:
+                missing.append(i)
+            else:
+                out[i] = vec
+        if missing:
+            fresh = self._embed([texts[i] for i in missing])
+            for i, vec in zip(missing, fresh):
+                out[i] = vec
+                self._cache.put(texts[i], vec)
+        return out
 
     @property
     def has_similarity(self) -> bool:

@@ -219,3 +219,27 @@ def test_agent_termination_frees_sessions():
         finally:
             eng.stop()
     asyncio.run(run())
+
+
+def test_embed_cache_hits_and_lru():
+    """SHA-keyed embedding cache with TTL + LRU (reference:
+    embeddings.ex:24-25,402-426)."""
+    from quoracle_amd.engine.fake import FakeEngine
+    from quoracle_amd.engine.pool import EmbedCache, EnginePool
+    eng = FakeEngine()
+    pool = EnginePool(default=eng, embedder=eng)
+    f = pool.embed_facade
+    a = f(["alpha", "beta"])
+    b = f(["alpha", "gamma"])
+    assert pool._embed_cache.hits == 1          # "alpha" served from cache
+    assert a[0] == b[0]
+    # repeated-batch call embeds only the misses
+    calls_before = len(eng.embed_calls)
+    f(["alpha", "beta", "gamma"])
+    assert len(eng.embed_calls) == calls_before  # all cached, no engine call
+    # LRU eviction under pressure
+    cache = EmbedCache(max_entries=10)
+    for i in range(12):
+        cache.put(f"t{i}", [float(i)])
+    assert cache.get("t0") is None               # oldest evicted
+    assert cache.get("t11") == [11.0]
