@@ -327,3 +327,43 @@ async def test_child_initial_message_carries_lineage_context():
     joined = "\n".join(str(e.get("content")) for e in h)
     assert "orient" in joined        # the parent's decision trail came along
     await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_announcement_reaches_all_descendants():
+    """send_message to='announcement' broadcasts a directive to every
+    descendant, depth-first (reference: actions/send_message.ex)."""
+    from quoracle_amd.actions import router as R
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("broadcast root", "default")
+    root_id = result["root_agent_id"]
+    root = runtime.registry.lookup(root_id).actor
+
+    # build a depth-2 tree through the real spawn path
+    spawn = {"task_description": "sub", "success_criteria": "s",
+             "immediate_context": "c", "approach_guidance": "a",
+             "profile": "default"}
+    r1 = await manager.supervisor.spawn_child_action(root, dict(spawn))
+    assert await wait_until(
+        lambda: runtime.registry.children_of(root_id), timeout=10)
+    child_id = runtime.registry.children_of(root_id)[0]
+    child = runtime.registry.lookup(child_id).actor
+    r2 = await manager.supervisor.spawn_child_action(child, dict(spawn))
+    assert await wait_until(
+        lambda: runtime.registry.children_of(child_id), timeout=10)
+    grand_id = runtime.registry.children_of(child_id)[0]
+
+    ctx = R.ActionContext(agent=root, runtime=runtime, action_id="a1",
+                          action="send_message",
+                          params={"to": "announcement",
+                                  "content": "all hands: freeze deploys"})
+    res = await R.execute_action(ctx)
+    assert set(res["delivered_to"]) == {child_id, grand_id}
+
+    def got(agent_id):
+        h = runtime.registry.lookup(agent_id).actor.state.model_histories
+        return any("freeze deploys" in str(e.get("content", ""))
+                   for e in h[POOL2[0]])
+    assert await wait_until(lambda: got(child_id) and got(grand_id), timeout=10)
+    await manager.supervisor.terminate_tree(root_id)
