@@ -136,6 +136,21 @@ class AgentActor:
                     continue
                 if mtype in MESSAGE_TYPES:
                     self.state.message_queue.append(message)
+                    # drain everything already waiting so simultaneous
+                    # messages batch into ONE cycle (reference:
+                    # message_batcher.ex; stale triggers/results folded in)
+                    while not self.inbox.empty():
+                        extra = self.inbox.get_nowait()
+                        etype = extra.get("type")
+                        if etype == "stop":
+                            self._running = False
+                            break
+                        if etype == "action_result":
+                            self._handle_action_result(extra)
+                        elif etype in MESSAGE_TYPES:
+                            self.state.message_queue.append(extra)
+                    if not self._running:
+                        break
                     await self._run_cycle()
                     continue
                 logger.warning("%s: unknown inbox message %s",
