@@ -231,10 +231,12 @@ class AgentActor:
         finally:
             self.state.consensus_in_flight = False
         if outcome is None:
+            self._flush_embedding_costs()
             self.state.status = "waiting"
             self.runtime.bus.state_change(self.state.agent_id, "waiting")
             return
         decision = outcome.decision
+        self._flush_embedding_costs()
         self.steps_completed += 1
         self.runtime.bus.consensus_decision(self.state.agent_id, {
             "kind": decision.kind,
@@ -306,13 +308,34 @@ class AgentActor:
             self.runtime.bus.task_message(self.state.task_id, {
                 "from": self.state.agent_id, "to": "user", "content": note})
 
+    EMBED_PRICE_PER_MTOK = 0.01    # matches the embed-small serving price
+
     def _embed_many(self):
         try:
-            # the facade is callable (embed_many) AND carries the fused
-            # similarity_matrix used by lesson dedup / clustering
-            return self.runtime.engines.embed_facade
+            # one facade per actor: callable embed_many + fused
+            # similarity_matrix + the per-cycle cost accumulator
+            if getattr(self, "_embed_facade", None) is None:
+                self._embed_facade = self.runtime.engines.embed_facade
+            return self._embed_facade
         except RuntimeError:
             return None
+
+    def _flush_embedding_costs(self) -> None:
+        """Flush the cycle's accumulated embedding cost in ONE record
+        (reference: message_handler.ex:133-144)."""
+        facade = getattr(self, "_embed_facade", None)
+        if facade is None:
+            return
+        tokens = facade.embedded_tokens - getattr(self, "_embed_flushed", 0)
+        if tokens <= 0:
+            return
+        self._embed_flushed = facade.embedded_tokens
+        cost = tokens * self.EMBED_PRICE_PER_MTOK / 1e6
+        self.state.budget_spent += cost
+        self.runtime.store.save_cost(
+            self.state.agent_id, self.state.task_id, "embed-small", cost,
+            category="embedding",
+            metadata={"tokens": tokens})
 
     def _last_user_prompt(self) -> str:
         for model in self.state.model_pool:

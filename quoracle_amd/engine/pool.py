@@ -105,6 +105,10 @@ class EmbedFacade:
         self._engine = engine
         self._embed = sync_embed_many(engine)
         self._cache = cache
+        # embedding-cost accumulator (reference: §2.6 — merged-rule
+        # embedding calls accumulate, the agent flushes once per cycle);
+        # cache hits are free, only fresh embeddings count
+        self.embedded_tokens = 0
 
     def __call__(self, texts: List[str]):
         texts = list(texts)
@@ -123,6 +127,8 @@ class EmbedFacade:
             for i, vec in zip(missing, fresh):
                 out[i] = vec
                 self._cache.put(texts[i], vec)
+            self.embedded_tokens += sum(
+                max(1, len(texts[i]) // 4) for i in missing)
         return out
 
     @property
@@ -130,6 +136,7 @@ class EmbedFacade:
         return hasattr(self._engine, "similarity_matrix")
 
     def similarity_matrix(self, texts: List[str]):
+        self.embedded_tokens += sum(max(1, len(t) // 4) for t in texts)
         if self.has_similarity:
             return self._engine.similarity_matrix(list(texts))
         from ..consensus.rules import cosine_similarity

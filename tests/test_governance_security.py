@@ -110,3 +110,37 @@ def test_vault_encryption_at_rest():
     assert raw is not None and b"supersecretvalue" not in raw
     assert v.get("tok") == "supersecretvalue"
     assert v.search(["to"]) == ["tok"]
+
+
+@pytest.mark.asyncio
+async def test_embedding_costs_flush_once_per_cycle():
+    """Embedding work during clustering/merging lands as ONE cost record
+    per cycle (reference: embedding-cost accumulator, §2.6)."""
+    import json as _json
+    from helpers import make_manager, wait_until, action_json
+    from quoracle_amd.engine.fake import FakeEngine
+
+    # semantic-rule action with diverging text -> embedding merges happen
+    # contents normalize to the same fingerprint (same keywords) but
+    # differ raw -> the cluster merges via the semantic rule = embeddings
+    variants = {"fake-a": "Status report: everything ready.",
+                "fake-b": "status REPORT... everything ready!!"}
+
+    def responder(model, msgs, req):
+        return action_json("send_message",
+                           {"to": "parent", "content": variants[model]})
+    engine = FakeEngine(response_fn=responder)
+    manager, runtime = make_manager(engine)
+    result = await manager.create_task("embed costs", "default")
+    root_id = result["root_agent_id"]
+    ok = await wait_until(
+        lambda: runtime.registry.lookup(root_id)
+        and runtime.registry.lookup(root_id).actor.steps_completed >= 1,
+        timeout=10)
+    assert ok
+    costs = runtime.store.costs_for_agent(root_id)
+    embed_rows = [c for c in costs if c["category"] == "embedding"]
+    assert embed_rows, "no embedding cost recorded"
+    steps = runtime.registry.lookup(root_id).actor.steps_completed
+    assert len(embed_rows) <= steps          # one flush per cycle max
+    assert embed_rows[0]["amount"] > 0
