@@ -31,6 +31,7 @@ from .tokenizer import EOS
 _SAFE_TEXT = [ord(c) for c in
               "abcdefghijklmnopqrstuvwxyz ABCDEFGHIJKLMNOPQRSTUVWXYZ"
               "0123456789.,:;!?()- "]
+_SAFE_SET = frozenset(_SAFE_TEXT)
 
 
 @dataclass
@@ -162,7 +163,12 @@ class ActionGrammar:
                 self.done = True
             return emitted
         if op[0] == FREE:
-            return sampled_id
+            # defensive: the engine masks sampling to _SAFE_TEXT, but any
+            # out-of-set id (misbehaving driver, replayed stream) must
+            # still yield valid JSON
+            if sampled_id in _SAFE_SET:
+                return sampled_id
+            return _SAFE_TEXT[sampled_id % len(_SAFE_TEXT)]
         # CHOICE: map the model's draw onto a candidate action
         action = self.candidates[sampled_id % len(self.candidates)]
         name_bytes = _encode(action)
