@@ -160,8 +160,12 @@ class AgentActor:
         except Exception:
             logger.exception("agent %s crashed", self.state.agent_id)
             self.runtime.bus.log(self.state.agent_id, "error", "agent crashed")
+            self._crashed = True
         finally:
             await self._terminate()
+            if getattr(self, "_crashed", False) and self.runtime.supervisor:
+                # supervision restart policy (reference: dyn_sup.ex 5/60s)
+                self.runtime.supervisor.schedule_restart(self.state.agent_id)
 
     async def _next_message(self) -> Optional[Dict[str, Any]]:
         """Block on the inbox honoring trigger/wait-timer semantics.

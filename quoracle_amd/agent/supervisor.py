@@ -317,6 +317,32 @@ class Supervisor:
         return {"status": "dismissing", "child_id": child_id}
 
     # -- restart policy -----------------------------------------------------------------
+    def schedule_restart(self, agent_id: str) -> None:
+        """Crash recovery: restore the agent from its persisted checkpoint
+        when the restart window allows (reference: DynamicSupervisor policy,
+        dyn_sup.ex:54-60); beyond the window the agent stays failed."""
+        if not self.record_crash(agent_id):
+            self.runtime.store.update_agent_status(agent_id, "failed")
+            self.runtime.bus.log(agent_id, "error",
+                                 "restart limit reached; agent failed")
+            return
+
+        async def _restart():
+            row = self.runtime.store.get_agent(agent_id)
+            if row is None or row.get("state") is None:
+                self.runtime.store.update_agent_status(agent_id, "failed")
+                return
+            try:
+                state = AgentState.from_checkpoint(row["state"])
+                self.start_agent(state)
+                self.runtime.bus.log(agent_id, "warning",
+                                     "agent restarted after crash")
+            except Exception:  # noqa: BLE001
+                logger.exception("restart of %s failed", agent_id)
+                self.runtime.store.update_agent_status(agent_id, "failed")
+
+        asyncio.ensure_future(_restart())
+
     def record_crash(self, agent_id: str) -> bool:
         """True if the agent may restart (max 5 restarts / 60 s, like the
         reference's DynamicSupervisor policy)."""
