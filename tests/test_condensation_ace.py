@@ -120,3 +120,38 @@ def test_lesson_dedup_via_fused_similarity_facade():
     # exact duplicate merged into base; the two identical new ones merged
     assert len(merged) == 2
     assert merged[0]["confidence"] >= 2 or merged[1]["confidence"] >= 2
+
+
+@pytest.mark.asyncio
+async def test_reflection_batched_by_context_budget():
+    """Large discarded spans reflect in multiple context-sized batches
+    (reference: condensation.ex:174-205)."""
+    st = _state(entries=0)
+    for i in range(12):
+        st.append_history(history_entry("event", f"note {i} " + "x" * 400))
+    eng = _reflective_engine()
+    eng._context_limits["fake-a"] = 1000   # batch budget ~400 tokens
+    ok = await cond.condense_model_history(st, "fake-a", eng)
+    assert ok
+    reflect_calls = [c for c in eng.calls]
+    assert len(reflect_calls) >= 2, "reflection was not batched"
+    marker = st.model_histories["fake-a"][-1]["content"]
+    assert "reflection batch" in marker
+
+
+@pytest.mark.asyncio
+async def test_oversized_entry_recursively_summarized():
+    """A single entry above 25% of the window is replaced by its recursive
+    summary (reference: condensation.ex:262-400)."""
+    st = _state(entries=1)
+    giant = "\n\n".join(f"paragraph {i}: " + "w" * 120 for i in range(40))
+    st.append_history(history_entry("event", giant))
+    eng = FakeEngine(default_response="condensed summary of the notes")
+    eng._context_limits["fake-a"] = 2000
+    changed = await cond.shrink_oversized_entries(st, "fake-a", eng)
+    assert changed
+    texts = [str(e.get("content", ""))
+             for e in st.model_histories["fake-a"]]
+    big = [t for t in texts if "oversized entry summarized" in t]
+    assert big
+    assert eng.count_tokens(big[0]) < eng.count_tokens(giant)
