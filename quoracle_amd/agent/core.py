@@ -441,11 +441,26 @@ class AgentActor:
             return result.text
         return query_fn
 
+    def _forbidden_actions(self) -> List[str]:
+        """Actions mechanically blocked by grove hard rules (reference:
+        consensus_handler.ex:294-333): excluded from prompts and grammar so
+        models never burn rounds proposing them."""
+        grove = self.state.grove or {}
+        out: List[str] = []
+        for rule in grove.get("hard_rules") or []:
+            if rule.get("type") == "action_block":
+                out.extend(a for a in rule.get("actions", [])
+                           if isinstance(a, str))
+        return out
+
     def _grammar_actions(self) -> List[str]:
         """Capability-gated action set for constrained decoding; the engine
         keeps the subset its grammar can template."""
         from ..governance import profiles as profiles_mod
-        return profiles_mod.allowed_actions(self.state.capability_groups)
+        forbidden = set(self._forbidden_actions())
+        return [a for a in
+                profiles_mod.allowed_actions(self.state.capability_groups)
+                if a not in forbidden]
 
     def _grammar_context(self) -> Dict[str, Any]:
         grove = self.state.grove or {}
@@ -488,6 +503,7 @@ class AgentActor:
             constraints=self.state.constraints,
             capability_groups=self.state.capability_groups
             if self.state.profile is not None else None,
+            forbidden_actions=self._forbidden_actions(),
             profile_names=self.runtime.profiles.names(),
             skills=self.state.active_skills,
             available_skills=available_skills,

@@ -128,6 +128,7 @@ def build_system_prompt(
     profile: Optional[profiles_mod.Profile] = None,
     constraints: Optional[List[str]] = None,
     capability_groups: Optional[List[str]] = None,
+    forbidden_actions: Optional[List[str]] = None,
     profile_names: Optional[List[str]] = None,
     skills: Optional[List[Dict[str, Any]]] = None,
     available_skills: Optional[List[Dict[str, str]]] = None,
@@ -172,6 +173,10 @@ def build_system_prompt(
     if caps is None and profile is not None:
         caps = profile.capability_groups
     available = profiles_mod.filter_actions(schema_mod.ACTIONS, caps)
+    if forbidden_actions:
+        # grove hard rules: mechanically blocked actions never appear in
+        # the offered schemas (reference: consensus_handler.ex:294-333)
+        available = [a for a in available if a not in set(forbidden_actions)]
     schema_docs = [format_action_schema(schema_mod.get_schema(a), profile_names)
                    for a in available]
     parts.append("## Available actions\n\n" + "\n\n".join(schema_docs))

@@ -144,3 +144,24 @@ async def test_embedding_costs_flush_once_per_cycle():
     steps = runtime.registry.lookup(root_id).actor.steps_completed
     assert len(embed_rows) <= steps          # one flush per cycle max
     assert embed_rows[0]["amount"] > 0
+
+
+@pytest.mark.asyncio
+async def test_grove_blocked_actions_excluded_from_prompt_and_grammar():
+    """Grove action_block rules remove the action from the offered schemas
+    AND the constrained-decoding candidates (reference:
+    consensus_handler.ex:294-333)."""
+    from helpers import make_manager, IDLE, wait_until
+    from quoracle_amd.engine.fake import FakeEngine
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    grove = {"name": "g", "path": "/tmp",
+             "hard_rules": [{"type": "action_block",
+                             "actions": ["fetch_web", "execute_shell"]}]}
+    result = await manager.create_task("blocked", "default", grove=grove)
+    root = runtime.registry.lookup(result["root_agent_id"]).actor
+    assert "fetch_web" not in root._grammar_actions()
+    prompt = root._system_prompt()
+    assert "fetch_web" not in prompt
+    assert "file_read" in prompt          # unblocked actions still offered
+    await manager.supervisor.terminate_tree(root.state.agent_id)
