@@ -381,6 +381,12 @@ class AgentActor:
                     correction=injectors.correction_block(correction_errors,
                                                           model_key),
                     refinement_prompt=refinement_prompt,
+                    # context-size telemetry block (reference:
+                    # consensus_handler/context_injector.ex)
+                    used_tokens=tm.history_tokens(
+                        engine.count_tokens,
+                        self.state.model_histories.get(model_key, [])),
+                    context_limit=engine.context_limit(model_key),
                 )
                 return [{"role": "system", "content": self._system_prompt()}] + convo
 
