@@ -197,26 +197,43 @@ class Sampler:
         Returns emitted token ids (grammar-adjusted)."""
         R = logits.shape[0]
         safe = _MaskCache.get(self.vocab_size, logits.device)
-        # Launch every row's sampling asynchronously (forced-token rows need
-        # no GPU work at all), then ONE host sync for the whole batch.
+        # Forced-token rows need no GPU work; the rest get ONE batched
+        # masked/temperature softmax, then a per-row multinomial (per-row
+        # generators keep per-request determinism) and ONE host sync.
         emitted: List[Optional[int]] = [None] * R
-        pending: List[Tuple[int, torch.Tensor]] = []
+        masked_rows: List[int] = []     # grammar FREE/CHOICE: printable only
+        plain_rows: List[int] = []      # unconstrained decode
         for r in range(R):
-            p, g = params[r], grammars[r]
+            g = grammars[r]
             if g is not None:
-                op = g.current()
-                if op[0] == FORCED:
+                if g.current()[0] == FORCED:
                     emitted[r] = g.advance(0)
-                    continue
-                row = logits[r] + safe        # FREE and CHOICE: printable only
+                else:
+                    masked_rows.append(r)
             else:
-                row = logits[r]
-            t = max(p.temperature, 1e-4)
-            probs = torch.softmax(row / t, dim=-1)
-            if p.top_p < 1.0:
-                probs = _top_p_filter(probs, p.top_p)
-            pending.append((r, torch.multinomial(probs, 1,
-                                                 generator=generators[r])))
+                plain_rows.append(r)
+
+        pending: List[Tuple[int, torch.Tensor]] = []
+
+        def _batch(rows: List[int], mask: Optional[torch.Tensor]) -> None:
+            if not rows:
+                return
+            temps = torch.tensor(
+                [max(params[r].temperature, 1e-4) for r in rows],
+                dtype=logits.dtype, device=logits.device).unsqueeze(1)
+            sub = logits[rows]
+            if mask is not None:
+                sub = sub + mask
+            probs = torch.softmax(sub / temps, dim=-1)
+            for i, r in enumerate(rows):
+                row_probs = probs[i]
+                if params[r].top_p < 1.0:
+                    row_probs = _top_p_filter(row_probs, params[r].top_p)
+                pending.append((r, torch.multinomial(
+                    row_probs, 1, generator=generators[r])))
+
+        _batch(masked_rows, safe)
+        _batch(plain_rows, None)
         if pending:
             ids = torch.cat([t for _, t in pending]).cpu()   # single sync
             for (r, _), idx in zip(pending, ids.tolist()):
