@@ -51,3 +51,22 @@ print(f"MFMA16 prefill: {ms_mfma:.3f} ms ({flops/ms_mfma/1e9:.1f} TFLOP/s)  "
       f"speedup {ms_valu/ms_mfma:.1f}x  rel-vs-valu {rel:.4f}")
 print(f"MFMA32 prefill: {ms_m32:.3f} ms ({flops/ms_m32/1e9:.1f} TFLOP/s)  "
       f"vs16 {ms_mfma/ms_m32:.2f}x  rel-vs-16 {rel32:.4f}")
+
+# --- decode path at bench-like shapes (B agents x 1 token, long ctx) -------
+B = 8
+ctx_len = 7000
+nb2 = (ctx_len + BS - 1) // BS + 1
+tables2 = torch.stack([torch.arange(1, nb2, dtype=torch.int32, device=dev)
+                       for _ in range(B)])
+qd = torch.randn(B, Hq, D, device=dev, dtype=torch.bfloat16)
+outd = torch.empty_like(qd)
+ctxs = torch.full((B,), ctx_len, dtype=torch.int32, device=dev)
+NSd = min(32, max(2, ctx_len // 256))
+pm = torch.empty((B, Hq, NSd), dtype=torch.float32, device=dev)
+pl = torch.empty_like(pm)
+pa = torch.empty((B, Hq, NSd, D), dtype=torch.float32, device=dev)
+ms_dec = timeit(lambda: ops.ext().paged_attn_decode_split(
+    outd, qd, kcache, vcache, tables2, ctxs, scale, pm, pl, pa), n=50)
+kv_bytes = 2 * B * ctx_len * Hkv * D * 2      # K+V bf16 read per call
+print(f"decode_split B={B} ctx={ctx_len} NS={NSd}: {ms_dec*1e3:.1f} us "
+      f"({kv_bytes/ms_dec/1e9:.2f} TB/s of {8} TB/s HBM peak)")
