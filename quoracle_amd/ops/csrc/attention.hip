@@ -392,13 +392,22 @@ paged_attn_decode_split_kernel(
     }
     __syncthreads();
 
+    // Phase B reads V from LDS, time-sharing k_s (K is fully consumed in
+    // phase A): one coalesced bulk load replaces 128 scattered row reads
+    // per chunk that measured 6% of HBM peak.
+    for (int i = tid; i < clen * dvecs; i += blockDim.x) {
+      const int key = i / dvecs, d8 = i % dvecs;
+      const int token = start + key;
+      const long blk = bt[(long)b * MAXB + token / BS];
+      const uint4 vv = reinterpret_cast<const uint4 *>(
+          vc + blk * panel_stride + ((long)hk * BS + token % BS) * D)[d8];
+      reinterpret_cast<uint4 *>(k_s + key * KSTRIDE + d8 * 8)[0] = vv;
+    }
+    __syncthreads();
+
     if (tid < D) {
       for (int i = 0; i < clen; ++i) {
-        const int token = start + i;
-        const long blk = bt[(long)b * MAXB + token / BS];
-        const bf16 *vrow =
-            vc + blk * panel_stride + ((long)hk * BS + token % BS) * D;
-        const float v = bf2f(vrow[tid]);
+        const float v = bf2f(k_s[i * KSTRIDE + tid]);
 #pragma unroll
         for (int g = 0; g < MAX_GQ; ++g) {
           if (g >= GQ) break;
