@@ -156,6 +156,14 @@ def create_app(manager) -> FastAPI:
                 "history_lengths": {m: len(h) for m, h in
                                     st.model_histories.items()}}
 
+    @app.get("/api/agents/{agent_id}/history/{kind}")
+    def agent_event_history(agent_id: str, kind: str):
+        """Ring-buffer replay (the reference's EventHistory mount replay):
+        kind in {logs, state, trace, todos, costs}."""
+        events = runtime.bus.history(f"agents:{agent_id}:{kind}")
+        return [{"type": e.type, "payload": e.payload, "ts": e.ts}
+                for e in events]
+
     # -- config surfaces (SecretManagementLive equivalents) ------------------
 
     @app.get("/api/profiles")

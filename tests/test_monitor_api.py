@@ -88,3 +88,17 @@ def test_prometheus_metrics(client):
     assert "quoracle_agents_alive" in body
     assert "quoracle_tasks_running 1" in body
     assert "quoracle_cost_usd_total" in body
+
+
+def test_event_history_replay(client):
+    c, runtime = client
+    r = c.post("/api/tasks", json={"prompt": "replay me",
+                                   "profile": "default"})
+    root = r.json()["root_agent_id"]
+    import time
+    deadline = time.monotonic() + 5
+    logs = []
+    while time.monotonic() < deadline and not logs:
+        logs = c.get(f"/api/agents/{root}/history/logs").json()
+        time.sleep(0.05)
+    assert logs and all("type" in e and "ts" in e for e in logs)
