@@ -243,3 +243,30 @@ def test_embed_cache_hits_and_lru():
         cache.put(f"t{i}", [float(i)])
     assert cache.get("t0") is None               # oldest evicted
     assert cache.get("t11") == [11.0]
+
+
+def test_multichunk_prefill_and_divergent_tail():
+    """Prompts longer than the prefill chunk stream in multiple chunks; a
+    session whose cached tail diverges mid-block re-prefills only from the
+    divergence point."""
+    eng = LocalEngine(["tiny#mc"], device=torch.device("cpu"),
+                      kv_blocks_override=512, embed_model_key=None,
+                      prefill_chunk=40)     # force several chunks
+    long_msg = "alpha " * 120               # ~600 tokens >> chunk of 40
+    r1 = eng.generate_sync(_req(model_key="tiny#mc", session_id="mc",
+                                messages=[{"role": "user",
+                                           "content": long_msg}]),
+                           timeout=240)
+    assert r1.ok
+    hm = eng.models["tiny#mc"]
+    sess = hm.sessions.get_or_create("mc")
+    cached = len(sess.token_ids)
+    assert cached > 3 * 40                  # several chunks landed
+    # divergent tail: same prefix, different ending
+    r2 = eng.generate_sync(_req(model_key="tiny#mc", session_id="mc",
+                                messages=[{"role": "user",
+                                           "content": long_msg + "OMEGA"}]),
+                           timeout=240)
+    assert r2.ok
+    hits = eng.stats["prefix_hit_tokens"]
+    assert hits > 0
