@@ -52,6 +52,14 @@ class TaskManager:
         output_style: Optional[str] = None,
         system_prompt: Optional[str] = None,
         task_id: Optional[str] = None,
+        # Task Work fields (reference: README "Create a Task" +
+        # fields/prompt_field_manager.ex): assembled into the first user
+        # prompt exactly like child spawns
+        success_criteria: Optional[str] = None,
+        immediate_context: Optional[str] = None,
+        approach_guidance: Optional[str] = None,
+        skills: Optional[List[str]] = None,
+        delegation_strategy: Optional[str] = None,
     ) -> Dict[str, Any]:
         try:
             profile = self.runtime.profiles.resolve(profile_name)
@@ -67,6 +75,22 @@ class TaskManager:
             "grove": grove,
         })
 
+        active_skills = []
+        if skills:
+            from ..governance.skills import SkillLoader, SkillError
+            loader = SkillLoader(self.runtime.config.skills_dir,
+                                 None if not grove or not grove.get("path")
+                                 else __import__("os").path.join(
+                                     grove["path"],
+                                     grove.get("skills_path") or "skills"))
+            for name in skills:
+                try:
+                    sk = loader.load(name)
+                    active_skills.append({"name": sk["name"],
+                                          "description": sk["description"],
+                                          "content": sk["content"]})
+                except SkillError:
+                    pass
         root_state = AgentState(
             agent_id=ids.agent_id("root"),
             task_id=task_id,
@@ -83,15 +107,25 @@ class TaskManager:
             grove=grove,
             system_prompt_fields={"system_prompt": system_prompt}
             if system_prompt else {},
+            active_skills=active_skills,
+            delegation_strategy=delegation_strategy,
             budget_mode="root" if budget_limit is not None else "na",
             budget_allocated=budget_limit,
         )
         root_state.init_model_maps()
         actor = self.supervisor.start_agent(root_state)
 
-        initial = prompt
+        parts = []
         if global_context:
-            initial = f"# Global context\n{global_context}\n\n# Task\n{prompt}"
+            parts.append(f"# Global context\n{global_context}")
+        parts.append(f"# Task\n{prompt}")
+        if success_criteria:
+            parts.append(f"# Success criteria\n{success_criteria}")
+        if immediate_context:
+            parts.append(f"# Immediate context\n{immediate_context}")
+        if approach_guidance:
+            parts.append(f"# Approach guidance\n{approach_guidance}")
+        initial = "\n\n".join(parts)
         await actor.deliver({"type": "user_message", "content": initial})
         return {"task_id": task_id, "root_agent_id": root_state.agent_id}
 

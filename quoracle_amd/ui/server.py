@@ -27,6 +27,13 @@ class CreateTaskBody(BaseModel):
     global_context: Optional[str] = None
     role: Optional[str] = None
     grove: Optional[str] = None      # grove name under the groves dir
+    success_criteria: Optional[str] = None
+    immediate_context: Optional[str] = None
+    approach_guidance: Optional[str] = None
+    skills: Optional[list] = None
+    cognitive_style: Optional[str] = None
+    output_style: Optional[str] = None
+    delegation_strategy: Optional[str] = None
 
 
 class MessageBody(BaseModel):
@@ -79,7 +86,12 @@ def create_app(manager) -> FastAPI:
             return await manager.create_task(
                 body.prompt, body.profile, budget_limit=body.budget_limit,
                 global_context=body.global_context, role=body.role,
-                grove=grove)
+                grove=grove, success_criteria=body.success_criteria,
+                immediate_context=body.immediate_context,
+                approach_guidance=body.approach_guidance,
+                skills=body.skills, cognitive_style=body.cognitive_style,
+                output_style=body.output_style,
+                delegation_strategy=body.delegation_strategy)
         except TaskError as exc:
             raise HTTPException(400, exc.reason)
 

@@ -102,3 +102,38 @@ def test_event_history_replay(client):
         logs = c.get(f"/api/agents/{root}/history/logs").json()
         time.sleep(0.05)
     assert logs and all("type" in e and "ts" in e for e in logs)
+
+
+def test_full_task_work_fields(client, tmp_path):
+    c, runtime = client
+    import os
+    skill_dir = tmp_path / "skills" / "analysis"
+    os.makedirs(skill_dir)
+    (skill_dir / "SKILL.md").write_text(
+        "---\nname: analysis\ndescription: analyze\n---\nLook closely.")
+    runtime.config.skills_dir = str(tmp_path / "skills")
+    r = c.post("/api/tasks", json={
+        "prompt": "build the report",
+        "profile": "default",
+        "success_criteria": "report.md exists",
+        "immediate_context": "data is in /data",
+        "approach_guidance": "start with the schema",
+        "skills": ["analysis"],
+        "cognitive_style": "systematic",
+        "delegation_strategy": "parallel"})
+    assert r.status_code == 200
+    root = r.json()["root_agent_id"]
+    actor = runtime.registry.lookup(root).actor
+    assert actor.state.cognitive_style == "systematic"
+    assert actor.state.delegation_strategy == "parallel"
+    assert any(s["name"] == "analysis" for s in actor.state.active_skills)
+    import time
+    deadline = time.monotonic() + 5
+    found = False
+    while time.monotonic() < deadline and not found:
+        h = actor.state.model_histories[actor.state.model_pool[0]]
+        found = any("Success criteria" in str(e.get("content", ""))
+                    and "Approach guidance" in str(e.get("content", ""))
+                    for e in h)
+        time.sleep(0.05)
+    assert found
