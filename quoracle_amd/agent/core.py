@@ -511,6 +511,7 @@ class AgentActor:
             if self.state.profile is not None else None,
             forbidden_actions=self._forbidden_actions(),
             profile_names=self.runtime.profiles.names(),
+            profile_catalog=self.runtime.profiles.catalog(),
             skills=self.state.active_skills,
             available_skills=available_skills,
             governance_docs=governance_docs,

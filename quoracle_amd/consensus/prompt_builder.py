@@ -130,6 +130,7 @@ def build_system_prompt(
     capability_groups: Optional[List[str]] = None,
     forbidden_actions: Optional[List[str]] = None,
     profile_names: Optional[List[str]] = None,
+    profile_catalog: Optional[List[Dict[str, Any]]] = None,
     skills: Optional[List[Dict[str, Any]]] = None,
     available_skills: Optional[List[Dict[str, str]]] = None,
     governance_docs: Optional[List[Dict[str, str]]] = None,
@@ -180,6 +181,12 @@ def build_system_prompt(
     schema_docs = [format_action_schema(schema_mod.get_schema(a), profile_names)
                    for a in available]
     parts.append("## Available actions\n\n" + "\n\n".join(schema_docs))
+    if profile_catalog and "spawn_child" in available:
+        lines = ["## Profiles available for child agents"]
+        for entry in profile_catalog:
+            desc = entry.get("description") or ""
+            lines.append(f"- {entry.get('name')}: {desc}".rstrip(": "))
+        parts.append("\n".join(lines))
 
     parts.append(RESPONSE_FORMAT)
     parts.append(EXAMPLE)

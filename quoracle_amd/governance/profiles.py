@@ -149,3 +149,10 @@ class ProfileStore:
 
     def names(self) -> List[str]:
         return list(self._profiles)
+
+    def catalog(self) -> List[Dict[str, str]]:
+        """name + description pairs — shown to parents when delegating
+        (reference: README 'profile names and descriptions are visible
+        to parent agents')."""
+        return [{"name": p.name, "description": p.description}
+                for p in self._profiles.values()]

@@ -165,3 +165,17 @@ async def test_grove_blocked_actions_excluded_from_prompt_and_grammar():
     assert "fetch_web" not in prompt
     assert "file_read" in prompt          # unblocked actions still offered
     await manager.supervisor.terminate_tree(root.state.agent_id)
+
+
+def test_profile_catalog_visible_to_spawners():
+    """Parents that can spawn see profile names + descriptions in their
+    system prompt; non-hierarchy agents don't get the section."""
+    from quoracle_amd.consensus.prompt_builder import build_system_prompt
+    catalog = [{"name": "researcher", "description": "read-only analysis"},
+               {"name": "builder", "description": "writes code"}]
+    spawner = build_system_prompt(capability_groups=["hierarchy"],
+                                  profile_catalog=catalog)
+    assert "read-only analysis" in spawner
+    leaf = build_system_prompt(capability_groups=[],
+                               profile_catalog=catalog)
+    assert "read-only analysis" not in leaf
