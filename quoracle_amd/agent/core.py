@@ -505,6 +505,7 @@ class AgentActor:
             role=self.state.role,
             cognitive_style=self.state.cognitive_style,
             output_style=self.state.output_style,
+            delegation_strategy=self.state.delegation_strategy,
             profile=profile,
             constraints=self.state.constraints,
             capability_groups=self.state.capability_groups

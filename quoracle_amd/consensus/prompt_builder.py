@@ -36,6 +36,14 @@ OUTPUT_STYLES = {
     "narrative": "Explain in flowing prose.",
 }
 
+DELEGATION_STRATEGIES = {
+    "sequential": "Delegate work to children one at a time, waiting for "
+                  "each result before spawning the next.",
+    "parallel": "Delegate independent work to several children at once "
+                "and integrate their results.",
+    "none": "Do the work yourself; do not spawn children.",
+}
+
 
 def _format_type(t: Any) -> str:
     if isinstance(t, str):
@@ -125,6 +133,7 @@ def build_system_prompt(
     role: Optional[str] = None,
     cognitive_style: Optional[str] = None,
     output_style: Optional[str] = None,
+    delegation_strategy: Optional[str] = None,
     profile: Optional[profiles_mod.Profile] = None,
     constraints: Optional[List[str]] = None,
     capability_groups: Optional[List[str]] = None,
@@ -151,6 +160,8 @@ def build_system_prompt(
         identity.append(COGNITIVE_STYLES[cognitive_style])
     if output_style in OUTPUT_STYLES:
         identity.append(OUTPUT_STYLES[output_style])
+    if delegation_strategy in DELEGATION_STRATEGIES:
+        identity.append(DELEGATION_STRATEGIES[delegation_strategy])
     parts.append("\n".join(identity))
 
     if extra_system_prompt:
