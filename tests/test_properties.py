@@ -98,3 +98,38 @@ def test_grammar_always_yields_valid_action_json(draws, allowed):
     parsed = json.loads(text)
     assert parsed["action"] in allowed
     assert isinstance(parsed["params"], dict)
+
+
+from quoracle_amd.actions import validator as V
+
+
+@given(st.dictionaries(st.sampled_from(["path", "mode", "content",
+                                        "old_string", "new_string",
+                                        "bogus_param"]),
+                       st.one_of(st.text(max_size=12), st.integers(),
+                                 st.booleans()),
+                       max_size=5))
+@settings(max_examples=80, deadline=None)
+def test_validator_never_crashes_and_gates_required(params):
+    """Arbitrary param dicts either validate or raise ValidationError —
+    never crash; file_write's required params stay enforced."""
+    try:
+        out = V.validate_params("file_write", dict(params))
+    except V.ValidationError as exc:
+        assert isinstance(exc.reason, str) and exc.reason
+        return
+    assert "path" in out and "mode" in out
+
+
+@given(st.lists(st.dictionaries(
+    st.sampled_from(["content", "state", "extra"]),
+    st.text(max_size=8), max_size=3), max_size=4))
+@settings(max_examples=60, deadline=None)
+def test_validator_todo_items_shape(items):
+    try:
+        out = V.validate_params("todo", {"items": items})
+    except V.ValidationError:
+        return
+    for item in out["items"]:
+        assert isinstance(item, dict)
+        assert item.get("state") in (None, "todo", "pending", "done")
