@@ -28,7 +28,19 @@ def entry_text(entry: Dict[str, Any]) -> str:
 
 
 def entry_tokens(count_tokens, entry: Dict[str, Any]) -> int:
-    return count_tokens(entry_text(entry))
+    """Token count of one history entry, memoized on the entry (entries are
+    append-only, so the count never goes stale; the byte tokenizer makes it
+    deterministic).  Keeps history_tokens O(new entries) per cycle instead
+    of O(history)."""
+    cached = entry.get("_tokens")
+    if isinstance(cached, int):
+        return cached
+    n = count_tokens(entry_text(entry))
+    try:
+        entry["_tokens"] = n
+    except TypeError:
+        pass
+    return n
 
 
 def history_tokens(count_tokens, history: List[Dict[str, Any]]) -> int:
