@@ -483,8 +483,11 @@ class LocalEngine(Engine):
                 torch.cuda.current_stream(self.device).wait_stream(hm.stream)
             logits = hm.graphs.run(
                 tokens, positions, slots, bt_rows, ctx_lens,
-                bt_keys=[(s.session.session_id, len(s.session.blocks),
-                          s.session.blocks[-1] if s.session.blocks else -1)
+                # the full block tuple participates in the staging key:
+                # after eviction + re-prefill a session can reuse the same
+                # length and tail block with different middle blocks
+                bt_keys=[(s.session.session_id,
+                          hash(tuple(s.session.blocks)))
                          for s in sample_seqs])
             if logits is not None and hm.stream is not None:
                 hm.stream.wait_stream(torch.cuda.current_stream(self.device))
