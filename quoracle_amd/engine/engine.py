@@ -627,6 +627,10 @@ class LocalEngine(Engine):
         for seq in done:
             if seq in hm.active:
                 hm.active.remove(seq)
+            # one-shot (anonymous) sessions release their KV immediately;
+            # named sessions persist for prefix reuse
+            if not seq.request.session_id and seq.session is not None:
+                hm.sessions.drop(seq.session.session_id)
             comp = seq._complete
             if comp is None:
                 continue

@@ -270,3 +270,14 @@ def test_multichunk_prefill_and_divergent_tail():
     assert r2.ok
     hits = eng.stats["prefix_hit_tokens"]
     assert hits > 0
+
+
+def test_anonymous_sessions_release_kv_on_finish():
+    eng = LocalEngine(["tiny#anon"], device=torch.device("cpu"),
+                      kv_blocks_override=128, embed_model_key=None,
+                      prefill_chunk=64)
+    free0 = eng.models["tiny#anon"].mgr.free_blocks
+    r = eng.generate_sync(_req(model_key="tiny#anon", session_id=""),
+                          timeout=120)
+    assert r.ok
+    assert eng.models["tiny#anon"].mgr.free_blocks == free0
