@@ -9,6 +9,10 @@ only synchronization is an admit-broadcast at the top of every engine step
 result futures; replicas execute identical forwards so the per-layer RCCL
 all-reduces line up (BASELINE config 5).
 
+Known limitation: the single-engine fault watchdog (engine.reset_model)
+is rank-local; a device fault on one TP rank desynchronizes the group and
+requires a group-level restart (tracked in ROADMAP.md).
+
 Exactness requirements (hold by construction):
   * identical model shards from the same seed (models/llama.py TP sharding),
   * identical KV block allocation (deterministic free-list),
