@@ -622,3 +622,218 @@ paged_attn_prefill_mfma32_kernel(
         f2bf(o_acc1[r] / denom);
   }
 }
+
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL (next-round validation): 64-row Q tiles, 8 waves.  4x K/V
+// staging reuse per query row vs the 16-row kernel.  Wave w: S-phase
+// keyblock w&3 for qblocks {2*(w>>2), 2*(w>>2)+1}; PV-phase d-columns
+// (w&3)*32..+31 for the same two qblocks.
+// Grid: (ntiles64, Hq); block 512.
+// ---------------------------------------------------------------------------
+#define MF4_QT 64
+
+extern "C" __global__ void __launch_bounds__(512)
+paged_attn_prefill_mfma64_kernel(
+    bf16 *__restrict__ out, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+
+  __shared__ bf16 q_s[MF4_QT * KP];
+  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ float s_s[MF4_QT * SP];
+  __shared__ bf16 p_s[MF4_QT * VP];
+  __shared__ float m_s[MF4_QT], l_s[MF4_QT], alpha_s[MF4_QT];
+
+  for (int i = tid; i < MF4_QT * MF_D / 8; i += 512) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
+  if (tid < MF4_QT) {
+    m_s[tid] = -INFINITY;
+    l_s[tid] = 0.f;
+  }
+  __syncthreads();
+
+  // two qblocks per wave: frags xN for qblock qb0 and qb0+1
+  f32x4_t s_acc0, s_acc1;
+  f32x4_t o_acc00 = {0.f, 0.f, 0.f, 0.f};   // qb0,   dcol kblock*32
+  f32x4_t o_acc01 = {0.f, 0.f, 0.f, 0.f};   // qb0,   dcol kblock*32+16
+  f32x4_t o_acc10 = {0.f, 0.f, 0.f, 0.f};   // qb0+1, dcol kblock*32
+  f32x4_t o_acc11 = {0.f, 0.f, 0.f, 0.f};   // qb0+1, dcol kblock*32+16
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int a_row = lane & 15;
+  const int a_koff = (lane >> 4) * 8;
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+  const int kblock = wave & 3;
+  const int qb0 = (wave >> 2) * 2;
+
+  uint4 kreg[2], vreg[2];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+    }
+  };
+
+  issue_loads(0, kv_limit);
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+    write_staged();
+    __syncthreads();
+    if (start + MF_KCHUNK < kv_limit)
+      issue_loads(start + MF_KCHUNK, kv_limit);
+
+    {
+      s_acc0 = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+      s_acc1 = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (kblock * 16 + a_row) * KP + kk * 32 + a_koff);
+        bf16x8_t a0 = *reinterpret_cast<const bf16x8_t *>(
+            q_s + ((qb0 + 0) * 16 + a_row) * KP + kk * 32 + a_koff);
+        bf16x8_t a1 = *reinterpret_cast<const bf16x8_t *>(
+            q_s + ((qb0 + 1) * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, s_acc0, 0, 0, 0);
+        s_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, s_acc1, 0, 0, 0);
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const f32x4_t &sa = half ? s_acc1 : s_acc0;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = (qb0 + half) * 16 + c_row0 + r;
+          const int col = kblock * 16 + c_col;
+          const int token = start + col;
+          const bool ok = (token <= pos0 + min(row, qn - 1)) && (col < clen);
+          s_s[row * SP + col] = ok ? sa[r] * scale : -INFINITY;
+        }
+      }
+    }
+    __syncthreads();
+
+    // softmax: 8 threads per row, 64 rows
+    {
+      const int row = tid >> 3;
+      const int sub = tid & 7;
+      float v[8];
+      float mymax = -INFINITY;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        v[j] = s_s[row * SP + sub + j * 8];
+        mymax = fmaxf(mymax, v[j]);
+      }
+#pragma unroll
+      for (int w = 4; w >= 1; w >>= 1)
+        mymax = fmaxf(mymax, __shfl_xor(mymax, w, 8));
+      const float m_old = m_s[row];
+      const float mn = fmaxf(m_old, mymax);
+      const float alpha = (m_old == -INFINITY) ? 0.f : __expf(m_old - mn);
+      float psum = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float p = (v[j] == -INFINITY || mn == -INFINITY)
+                            ? 0.f : __expf(v[j] - mn);
+        p_s[row * VP + sub + j * 8] = f2bf(p);
+        psum += p;
+      }
+#pragma unroll
+      for (int w = 4; w >= 1; w >>= 1)
+        psum += __shfl_xor(psum, w, 8);
+      if (sub == 0) {
+        l_s[row] = l_s[row] * alpha + psum;
+        m_s[row] = mn;
+        alpha_s[row] = alpha;
+      }
+    }
+    __syncthreads();
+
+    {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float a0 = alpha_s[(qb0 + 0) * 16 + c_row0 + r];
+        const float a1 = alpha_s[(qb0 + 1) * 16 + c_row0 + r];
+        o_acc00[r] *= a0;
+        o_acc01[r] *= a0;
+        o_acc10[r] *= a1;
+        o_acc11[r] *= a1;
+      }
+#pragma unroll
+      for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
+        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (kblock * 32 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (kblock * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t a0 = *reinterpret_cast<const bf16x8_t *>(
+            p_s + ((qb0 + 0) * 16 + a_row) * VP + kk * 32 + a_koff);
+        bf16x8_t a1 = *reinterpret_cast<const bf16x8_t *>(
+            p_s + ((qb0 + 1) * 16 + a_row) * VP + kk * 32 + a_koff);
+        o_acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, o_acc00, 0, 0, 0);
+        o_acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, o_acc01, 0, 0, 0);
+        o_acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, o_acc10, 0, 0, 0);
+        o_acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, o_acc11, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = (qb0 + half) * 16 + c_row0 + r;
+      if (row >= qn) continue;
+      const float denom = l_s[row] > 0.f ? l_s[row] : 1.f;
+      const f32x4_t &oa = half ? o_acc10 : o_acc00;
+      const f32x4_t &ob = half ? o_acc11 : o_acc01;
+      out[((long)(q0 + row) * Hq + h) * MF_D + kblock * 32 + c_col] =
+          f2bf(oa[r] / denom);
+      out[((long)(q0 + row) * Hq + h) * MF_D + kblock * 32 + 16 + c_col] =
+          f2bf(ob[r] / denom);
+    }
+  }
+}

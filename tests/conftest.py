@@ -16,6 +16,11 @@ def pytest_configure(config):
     config.addinivalue_line(
         "markers", "asyncio: run the test function in a fresh asyncio event loop"
     )
+    config.addinivalue_line(
+        "markers", "gpu_experimental: unvalidated GPU kernels — run "
+        "explicitly with -m gpu_experimental on a GPU box; excluded from "
+        "the standard gpu suite"
+    )
 
 
 @pytest.hookimpl(tryfirst=True)
@@ -41,5 +46,5 @@ def pytest_collection_modifyitems(config, items):
         return
     skip_gpu = pytest.mark.skip(reason="no GPU available")
     for item in items:
-        if "gpu" in item.keywords:
+        if "gpu" in item.keywords or "gpu_experimental" in item.keywords:
             item.add_marker(skip_gpu)
