@@ -650,3 +650,152 @@ paged_attn_prefill_reduce_kernel(bf16 *__restrict__ out,
     __syncthreads();
   }
 }
+
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL (next-round validation): decode split v2 — 256-token chunks
+// with 256 threads (half the barrier count per KV token), chunk block-ids
+// precomputed once into LDS, K and V staged through LDS as in v1.
+//   part layout identical to paged_attn_decode_split_kernel; reuses the
+//   same reduce kernel.
+// Grid: (B, Hkv, NS); block 256.  Requires DCHUNK % BS == 0.
+// ---------------------------------------------------------------------------
+#define DCHUNK2 256
+#define DBLOCK2 256
+
+extern "C" __global__ void __launch_bounds__(DBLOCK2)
+paged_attn_decode_split2_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ ctx, float scale,
+    int Hq, int Hkv, int D, int BS, int MAXB, int GQ, int NS) {
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int split = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int len = ctx[b];
+  const int chunks = (len + DCHUNK2 - 1) / DCHUNK2;
+  const int per = (chunks + NS - 1) / NS;
+  const int t0 = split * per * DCHUNK2;
+  const int t1 = min(len, (split + 1) * per * DCHUNK2);
+  const bool dead = (t0 >= len);
+
+  __shared__ float q_s[MAX_GQ * 128];
+  __shared__ float p_s[MAX_GQ][DCHUNK2];
+  __shared__ bf16 k_s[DCHUNK2 * KSTRIDE];
+  __shared__ long blk_s[DCHUNK2 / 16 + 1];   // BS >= 16 -> <= 17 blocks
+  __shared__ float scratch[8];
+
+  for (int i = tid; i < GQ * D; i += blockDim.x) {
+    int g = i / D, d = i % D;
+    q_s[g * D + d] = bf2f(q[((long)b * Hq + hk * GQ + g) * D + d]) * scale;
+  }
+
+  float m[MAX_GQ], l[MAX_GQ], acc[MAX_GQ];
+#pragma unroll
+  for (int g = 0; g < MAX_GQ; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+    acc[g] = 0.f;
+  }
+
+  const long panel_stride = (long)Hkv * BS * D;
+  const int dvecs = D / 8;
+  const int blocks_per_chunk = DCHUNK2 / BS;
+  for (int start = t0; start < t1; start += DCHUNK2) {
+    const int clen = min(DCHUNK2, t1 - start);
+    // chunk block ids once (start is DCHUNK2-aligned and DCHUNK2 % BS == 0)
+    if (tid <= (clen - 1) / BS)
+      blk_s[tid] = bt[(long)b * MAXB + start / BS + tid];
+    __syncthreads();
+
+    for (int i = tid; i < clen * dvecs; i += blockDim.x) {
+      const int key = i / dvecs, d8 = i % dvecs;
+      const long blk = blk_s[key / BS];
+      const uint4 kv = reinterpret_cast<const uint4 *>(
+          kc + blk * panel_stride + ((long)hk * BS + (start + key) % BS) * D)[d8];
+      reinterpret_cast<uint4 *>(k_s + key * KSTRIDE + d8 * 8)[0] = kv;
+    }
+    __syncthreads();
+
+    if (tid < clen) {
+      float dots[MAX_GQ];
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) dots[g] = 0.f;
+      const bf16 *krow = k_s + tid * KSTRIDE;
+      for (int d8 = 0; d8 < dvecs; ++d8) {
+        uint4 kv = reinterpret_cast<const uint4 *>(krow)[d8];
+        float kf[8];
+        unpack_bf16x2(kv.x, kf[0], kf[1]);
+        unpack_bf16x2(kv.y, kf[2], kf[3]);
+        unpack_bf16x2(kv.z, kf[4], kf[5]);
+        unpack_bf16x2(kv.w, kf[6], kf[7]);
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          const float *qg = q_s + g * D + d8 * 8;
+#pragma unroll
+          for (int kx = 0; kx < 8; ++kx)
+            dots[g] = fmaf(kf[kx], qg[kx], dots[g]);
+        }
+      }
+#pragma unroll
+      for (int g = 0; g < MAX_GQ; ++g) {
+        if (g >= GQ) break;
+        p_s[g][tid] = dots[g];
+      }
+    }
+    __syncthreads();
+
+    for (int g = 0; g < GQ; ++g) {
+      float mine = (tid < clen) ? p_s[g][tid] : -INFINITY;
+      float cmax = block_max(mine, scratch);
+      float mn = fmaxf(m[g], cmax);
+      float alpha = (m[g] == -INFINITY) ? 0.f : __expf(m[g] - mn);
+      float p = (tid < clen && mn != -INFINITY)
+                    ? __expf(p_s[g][tid] - mn) : 0.f;
+      if (tid < clen) p_s[g][tid] = p;
+      float psum = block_sum(p, scratch);
+      l[g] = l[g] * alpha + psum;
+      acc[g] *= alpha;
+      m[g] = mn;
+    }
+    __syncthreads();
+
+    // V through LDS (time-shares k_s)
+    for (int i = tid; i < clen * dvecs; i += blockDim.x) {
+      const int key = i / dvecs, d8 = i % dvecs;
+      const long blk = blk_s[key / BS];
+      const uint4 vv = reinterpret_cast<const uint4 *>(
+          vc + blk * panel_stride + ((long)hk * BS + (start + key) % BS) * D)[d8];
+      reinterpret_cast<uint4 *>(k_s + key * KSTRIDE + d8 * 8)[0] = vv;
+    }
+    __syncthreads();
+
+    if (tid < D) {
+      for (int i = 0; i < clen; ++i) {
+        const float v = bf2f(k_s[i * KSTRIDE + tid]);
+#pragma unroll
+        for (int g = 0; g < MAX_GQ; ++g) {
+          if (g >= GQ) break;
+          acc[g] = fmaf(p_s[g][i], v, acc[g]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  if (tid < D) {
+#pragma unroll
+    for (int g = 0; g < MAX_GQ; ++g) {
+      if (g >= GQ) break;
+      const int h = hk * GQ + g;
+      const long base = ((long)b * Hq + h) * NS + split;
+      if (tid == 0) {
+        part_m[base] = dead ? -INFINITY : m[g];
+        part_l[base] = dead ? 0.f : l[g];
+      }
+      part_acc[base * D + tid] = dead ? 0.f : acc[g];
+    }
+  }
+}

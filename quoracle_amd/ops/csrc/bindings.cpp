@@ -175,6 +175,45 @@ void paged_attn_decode_split(torch::Tensor out, torch::Tensor q,
   HIP_CHECK_KERNEL();
 }
 
+void paged_attn_decode_split2(torch::Tensor out, torch::Tensor q,
+                              torch::Tensor kcache, torch::Tensor vcache,
+                              torch::Tensor block_tables,
+                              torch::Tensor ctx_lens, double scale,
+                              torch::Tensor part_m, torch::Tensor part_l,
+                              torch::Tensor part_acc) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  CHECK_GPU(ctx_lens);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  const int NS = part_m.size(2);
+  TORCH_CHECK(Hq % Hkv == 0 && GQ <= MAX_GQ, "unsupported GQA ratio");
+  TORCH_CHECK(D <= 128, "head dim too large");
+  TORCH_CHECK(256 % BS == 0 && BS <= 256, "block size must divide 256");
+  if (B == 0) return;
+  hipLaunchKernelGGL(paged_attn_decode_split2_kernel, dim3(B, Hkv, NS),
+                     dim3(256), 0, current_stream(),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), bf16_cptr(q),
+                     bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
+                     (float)scale, Hq, Hkv, D, BS, MAXB, GQ, NS);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(paged_attn_decode_reduce_kernel, dim3(B, Hq),
+                     dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), Hq, D, NS);
+  HIP_CHECK_KERNEL();
+}
+
 void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
                         torch::Tensor kcache, torch::Tensor vcache,
                         torch::Tensor block_tables, torch::Tensor tile_q0,
@@ -426,6 +465,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_append", &kv_append, "Scatter K/V rows into the paged cache");
   m.def("paged_attn_decode", &paged_attn_decode,
         "Paged-KV GQA decode attention (one token per sequence)");
+  m.def("paged_attn_decode_split2", &paged_attn_decode_split2,
+        "EXPERIMENTAL: decode split v2 (256-token chunks, LDS block ids)");
   m.def("paged_attn_decode_split", &paged_attn_decode_split,
         "Flash-decoding: context-split decode attention + combine");
   m.def("paged_attn_prefill", &paged_attn_prefill,
