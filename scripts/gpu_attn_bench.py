@@ -70,3 +70,24 @@ ms_dec = timeit(lambda: ops.ext().paged_attn_decode_split(
 kv_bytes = 2 * B * ctx_len * Hkv * D * 2      # K+V bf16 read per call
 print(f"decode_split B={B} ctx={ctx_len} NS={NSd}: {ms_dec*1e3:.1f} us "
       f"({kv_bytes/ms_dec/1e9:.2f} TB/s of {8} TB/s HBM peak)")
+
+# --- experimental kernels (validate with tests/test_ops_experimental.py
+# BEFORE trusting these numbers) --------------------------------------------
+if hasattr(ops.ext(), "paged_attn_prefill_mfma64"):
+    nt64 = (new + 63) // 64
+    t64 = torch.arange(nt64, dtype=torch.int32, device=dev) * 64
+    qn64 = torch.clamp(torch.full_like(t64, new) - t64, max=64)
+    ts64 = torch.zeros_like(t64)
+    tp64 = t64 + cached
+    ms_m64 = timeit(lambda: ops.ext().paged_attn_prefill_mfma64(
+        out, q, kcache, vcache, tables, t64, qn64, ts64, tp64, scale))
+    rel64 = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
+    print(f"MFMA64 prefill (EXPERIMENTAL): {ms_m64:.3f} ms "
+          f"({flops/ms_m64/1e9:.1f} TFLOP/s)  rel-vs-16 {rel64:.4f}")
+if hasattr(ops.ext(), "paged_attn_decode_split2"):
+    outd2 = torch.empty_like(qd)
+    ms_dec2 = timeit(lambda: ops.ext().paged_attn_decode_split2(
+        outd2, qd, kcache, vcache, tables2, ctxs, scale, pm, pl, pa), n=50)
+    reld = (outd2.float() - outd.float()).norm() / outd.float().norm()
+    print(f"decode_split2 (EXPERIMENTAL): {ms_dec2*1e3:.1f} us "
+          f"({kv_bytes/ms_dec2/1e9:.2f} TB/s)  rel-vs-v1 {reld:.4f}")
