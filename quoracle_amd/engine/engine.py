@@ -513,6 +513,7 @@ class LocalEngine(Engine):
 
         row = n_decode
         max_kv = 0
+        big_step = 64 if os.environ.get("QUORACLE_MFMA64") else 32
         for seq, n in prefill:
             cached = len(seq.session.token_ids)
             chunk = seq.known[cached:cached + n]
@@ -529,9 +530,9 @@ class LocalEngine(Engine):
                 tile_qn.append(min(QT, n - off))
                 tile_seq.append(seq_row)
                 tile_pos0.append(cached + off)
-            for off in range(0, n, 2 * QT):
+            for off in range(0, n, big_step):
                 t32_q0.append(row + off)
-                t32_qn.append(min(2 * QT, n - off))
+                t32_qn.append(min(big_step, n - off))
                 t32_seq.append(seq_row)
                 t32_pos0.append(cached + off)
             if cached + n == len(seq.known):

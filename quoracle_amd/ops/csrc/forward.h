@@ -13,6 +13,7 @@
 
 #pragma once
 
+#include <cstdlib>
 #include <unordered_map>
 #include <vector>
 
@@ -99,8 +100,13 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
       auto pm = torch::empty({B, Hq, ns}, opts);
       auto pl = torch::empty({B, Hq, ns}, opts);
       auto pa = torch::empty({B, Hq, ns, D}, opts);
-      paged_attn_decode_split(out_d, q_d, kcache, vcache, block_tables,
-                              ctx_lens.value(), m.scale, pm, pl, pa);
+      static const bool dec2 = std::getenv("QUORACLE_DECODE_V2") != nullptr;
+      if (dec2)
+        paged_attn_decode_split2(out_d, q_d, kcache, vcache, block_tables,
+                                 ctx_lens.value(), m.scale, pm, pl, pa);
+      else
+        paged_attn_decode_split(out_d, q_d, kcache, vcache, block_tables,
+                                ctx_lens.value(), m.scale, pm, pl, pa);
     } else {
       paged_attn_decode(out_d, q_d, kcache, vcache, block_tables,
                         ctx_lens.value(), m.scale);
@@ -124,10 +130,18 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
                                       *tile_seq, *tile_pos0, m.scale, pm, pl,
                                       pa);
       } else if (t32_q0.has_value() && t32_q0->numel() > 0) {
-        // big prefill: 8-wave 32-row tiles (2x K/V reuse per query row)
-        paged_attn_prefill_mfma32(attn_out, q, kcache, vcache, block_tables,
-                                  *t32_q0, *t32_qn, *t32_seq, *t32_pos0,
-                                  m.scale);
+        // big prefill: 8-wave big tiles (2-4x K/V reuse per query row);
+        // QUORACLE_MFMA64 promotes the experimental 64-row variant (the
+        // engine sizes the tiles to match)
+        static const bool use64 = std::getenv("QUORACLE_MFMA64") != nullptr;
+        if (use64)
+          paged_attn_prefill_mfma64(attn_out, q, kcache, vcache,
+                                    block_tables, *t32_q0, *t32_qn,
+                                    *t32_seq, *t32_pos0, m.scale);
+        else
+          paged_attn_prefill_mfma32(attn_out, q, kcache, vcache,
+                                    block_tables, *t32_q0, *t32_qn,
+                                    *t32_seq, *t32_pos0, m.scale);
       } else {
         paged_attn_prefill_mfma(attn_out, q, kcache, vcache, block_tables,
                                 *tile_q0, *tile_qn, *tile_seq, *tile_pos0,
