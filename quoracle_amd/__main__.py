@@ -88,6 +88,13 @@ def main() -> None:
         "quoracle_amd.tools.show_prompts", fromlist=["main"]).main(
             [a.scenario]))
 
+    stk = sub.add_parser("show-task", help="render a task transcript")
+    stk.add_argument("db")
+    stk.add_argument("task_id", nargs="?")
+    stk.set_defaults(fn=lambda a: __import__(
+        "quoracle_amd.tools.show_task", fromlist=["main"]).main(
+            [a.db] + ([a.task_id] if a.task_id else [])))
+
     args = p.parse_args()
     args.fn(args)
 

@@ -166,3 +166,26 @@ async def test_generate_images_without_model_is_structured():
                                       {"prompt": "a red square"}))
     assert isinstance(res, dict)
     assert res.get("error") or res.get("status")
+
+
+@pytest.mark.asyncio
+async def test_show_task_transcript(tmp_path):
+    """show-task renders the persisted tree, logs, costs and messages."""
+    from quoracle_amd.persistence.store import Store
+    from quoracle_amd.tools.show_task import render_task
+    from helpers import make_manager, wait_until
+    store = Store(str(tmp_path / "t.db"))
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine, store=store)
+    result = await manager.create_task("transcribe me", "default")
+    root_id = result["root_agent_id"]
+    assert await wait_until(
+        lambda: runtime.registry.lookup(root_id)
+        and runtime.registry.lookup(root_id).actor.steps_completed >= 1,
+        timeout=10)
+    await manager.send_user_message(result["task_id"], "hello transcript")
+    await manager.supervisor.terminate_tree(root_id)
+    text = render_task(store, result["task_id"])
+    assert "transcribe me" in text
+    assert root_id in text
+    assert "Agent tree" in text and "Messages" in text
