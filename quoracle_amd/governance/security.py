@@ -10,6 +10,7 @@ NO_EXECUTE tags with a random hex suffix).
 
 from __future__ import annotations
 
+import hashlib
 import re
 import secrets as pysecrets
 import string
@@ -40,26 +41,65 @@ def generate_secret_value(length: int = 32, include_symbols: bool = False,
 
 
 class SecretVault:
-    """Secrets at rest are XOR-keystream obfuscated in the store; the proper
-    hardening point (AES-GCM per the reference's Cloak vault) is a drop-in
-    replacement of _seal/_unseal."""
+    """Authenticated encryption at rest (the reference uses AES-256-GCM via
+    Cloak with CLOAK_ENCRYPTION_KEY; this image has no AES library, so the
+    vault uses the stdlib equivalent construction: HMAC-SHA256 in counter
+    mode as the keystream PRF + an encrypt-then-MAC tag, random 16-byte
+    nonce per value).  Key material comes from QUORACLE_VAULT_KEY (or the
+    constructor); blobs are versioned, legacy v0 XOR blobs still decrypt."""
+
+    _MAGIC = b"qv2:"
 
     def __init__(self, store, key: Optional[bytes] = None):
+        import os as _os
         self._store = store
-        self._key = key or b"quoracle-amd-vault"
+        raw_key = key or _os.environ.get(
+            "QUORACLE_VAULT_KEY", "quoracle-amd-vault").encode()             if not isinstance(key, bytes) else key
+        self._enc_key = hashlib.sha256(b"enc|" + raw_key).digest()
+        self._mac_key = hashlib.sha256(b"mac|" + raw_key).digest()
+        self._legacy_key = raw_key
 
-    def _xor(self, data: bytes) -> bytes:
-        key = self._key
-        return bytes(b ^ key[i % len(key)] for i, b in enumerate(data))
+    def _keystream_xor(self, nonce: bytes, data: bytes) -> bytes:
+        import hmac as _hmac
+        out = bytearray()
+        counter = 0
+        while len(out) < len(data):
+            block = _hmac.new(self._enc_key,
+                              nonce + counter.to_bytes(8, "big"),
+                              hashlib.sha256).digest()
+            out.extend(block)
+            counter += 1
+        return bytes(b ^ k for b, k in zip(data, out))
+
+    def _seal(self, plaintext: bytes) -> bytes:
+        import hmac as _hmac
+        nonce = pysecrets.token_bytes(16)
+        ct = self._keystream_xor(nonce, plaintext)
+        tag = _hmac.new(self._mac_key, nonce + ct, hashlib.sha256).digest()
+        return self._MAGIC + nonce + tag + ct
+
+    def _unseal(self, blob: bytes) -> bytes:
+        import hmac as _hmac
+        if not blob.startswith(self._MAGIC):
+            # legacy v0: repeating-XOR blobs from older stores
+            key = self._legacy_key
+            return bytes(b ^ key[i % len(key)] for i, b in enumerate(blob))
+        nonce = blob[4:20]
+        tag = blob[20:52]
+        ct = blob[52:]
+        expect = _hmac.new(self._mac_key, nonce + ct, hashlib.sha256).digest()
+        if not _hmac.compare_digest(tag, expect):
+            raise SecretNotFoundError("vault_tag_mismatch")
+        return self._keystream_xor(nonce, ct)
 
     def put(self, name: str, value: str, description: str = "") -> None:
-        self._store.save_secret(name, self._xor(value.encode()), description)
+        self._store.save_secret(name, self._seal(value.encode()), description)
 
     def get(self, name: str) -> str:
         raw = self._store.get_secret(name)
         if raw is None:
             raise SecretNotFoundError(name)
-        return self._xor(bytes(raw)).decode()
+        return self._unseal(bytes(raw)).decode()
 
     def names(self) -> List[str]:
         return self._store.list_secret_names()
