@@ -53,8 +53,13 @@ class SecretVault:
     def __init__(self, store, key: Optional[bytes] = None):
         import os as _os
         self._store = store
-        raw_key = key or _os.environ.get(
-            "QUORACLE_VAULT_KEY", "quoracle-amd-vault").encode()             if not isinstance(key, bytes) else key
+        if key is None:
+            raw_key = _os.environ.get("QUORACLE_VAULT_KEY",
+                                      "quoracle-amd-vault").encode()
+        elif isinstance(key, str):
+            raw_key = key.encode()
+        else:
+            raw_key = key
         self._enc_key = hashlib.sha256(b"enc|" + raw_key).digest()
         self._mac_key = hashlib.sha256(b"mac|" + raw_key).digest()
         self._legacy_key = raw_key
