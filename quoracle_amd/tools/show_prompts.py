@@ -24,6 +24,7 @@ SCENARIOS = [
     "initial", "with_profile", "with_constraints", "with_skills",
     "with_governance", "conversation", "injectors", "refinement",
     "final_round", "correction", "orient_result", "condensed",
+    "constrained_decode",
 ]
 
 
@@ -97,6 +98,22 @@ def render_scenario(name: str) -> str:
             {"llama3-8b#0": "invalid_json"}, "llama3-8b#0")
     if name == "orient_result":
         return json.dumps(_sample_history()[1]["content"], indent=2)
+    if name == "constrained_decode":
+        # deterministic grammar walk: what the engine's constrained decoder
+        # emits for a fixed stream of sampler draws
+        from ..engine.sampler import ActionGrammar
+        from ..engine.tokenizer import EOS, ByteTokenizer
+        g = ActionGrammar(["orient", "send_message", "todo", "wait"],
+                          reasoning_tokens=8,
+                          context={"spawn_profile": "default"})
+        draws = [ord(c) for c in "assessing the task now then act"]
+        out, i = [], 0
+        while not g.done and len(out) < 2000:
+            out.append(g.advance(draws[i % len(draws)]))
+            i += 1
+        text = ByteTokenizer().decode([t for t in out if t != EOS])
+        return (f"sampler draws: {draws[:8]}... (cycled)\n"
+                f"emitted action JSON ({len(out)} tokens):\n{text}")
     if name == "condensed":
         from ..agent import condensation as cond
         fn = getattr(cond, "condensation_artifact", None)
