@@ -412,7 +412,12 @@ class AgentActor:
                 # of the unchanged history prefix (engine prefix cache)
                 session_id=f"{self.state.agent_id}:{model_key}",
             )
-            result = await engine.generate(request)
+            try:
+                result = await asyncio.wait_for(
+                    engine.generate(request),
+                    timeout=self.runtime.config.generate_timeout_s)
+            except asyncio.TimeoutError:
+                raise RuntimeError("generate_timeout") from None
             if self.runtime.config.trace_prompts:
                 self.runtime.bus.broadcast(
                     f"agents:{self.state.agent_id}:trace", "llm_exchange", {
@@ -427,8 +432,10 @@ class AgentActor:
                 # condense once and retry (reference: per_model_query.ex:93-124)
                 await condensation_mod.condense_model_history(
                     self.state, model_key, engine, embed_many=self._embed_many())
-                result = await engine.generate(
-                    GenerateRequest(**{**request.__dict__, "messages": _build()}))
+                result = await asyncio.wait_for(
+                    engine.generate(GenerateRequest(
+                        **{**request.__dict__, "messages": _build()})),
+                    timeout=self.runtime.config.generate_timeout_s)
             if not result.ok:
                 raise RuntimeError(result.error or "query_failed")
             if result.cost:

@@ -34,6 +34,9 @@ class RuntimeConfig:
     shell_timeout_s: float = 600.0
     spawn_retries: int = 3
     consensus_retries: int = 3            # ref: message_handler.ex:353-421
+    # hard cap on one model generate (a wedged engine degrades to a
+    # per-model failure -> partial-pool consensus instead of a hung agent)
+    generate_timeout_s: float = 600.0
     test_mode: bool = False
     # verbose prompt tracing: broadcast every sent message list + raw
     # response on agents:<id>:trace (reference: consensus_handler.ex:154-179

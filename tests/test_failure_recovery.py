@@ -181,3 +181,30 @@ async def test_restart_limit_marks_agent_failed():
                      .get("status") == "failed", timeout=5)
     assert runtime.store.get_agent(root_id)["status"] == "failed"
     assert not runtime.registry.alive(root_id)
+
+
+@pytest.mark.asyncio
+async def test_wedged_engine_degrades_to_partial_pool():
+    """A model whose engine never answers times out and drops out of the
+    vote; the cycle still decides via the responsive model."""
+    import asyncio as _a
+
+    class WedgedOnA(FakeEngine):
+        async def generate(self, request):
+            if request.model_key == "fake-a":
+                await _a.sleep(30)          # never within the timeout
+            return await super().generate(request)
+
+    from quoracle_amd.tasks.runtime import RuntimeConfig
+    from helpers import make_manager, action_json, wait_until
+    engine = WedgedOnA(default_response=action_json(
+        "todo", {"items": [{"content": "w", "state": "todo"}]}))
+    manager, runtime = make_manager(
+        engine, config=RuntimeConfig(generate_timeout_s=0.3))
+    result = await manager.create_task("wedge", "default")
+    root_id = result["root_agent_id"]
+    ok = await wait_until(
+        lambda: runtime.registry.lookup(root_id)
+        and runtime.registry.lookup(root_id).actor.state.todos, timeout=15)
+    assert ok, "agent hung on the wedged engine"
+    await manager.supervisor.terminate_tree(root_id)
