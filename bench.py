@@ -316,6 +316,7 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
             "decisions_completed": decisions,
             "constrained_decoding": True,
             "max_refinement_rounds": 2,
+            "engine_stats": dict(engine.stats),
         },
     }
 
