@@ -140,6 +140,9 @@ class LocalEngine(Engine):
                       "requests_done": 0, "prefix_hit_tokens": 0}
         self._log_every = int(os.environ.get("QUORACLE_ENGINE_LOG", "0"))
         self._roctx = bool(os.environ.get("QUORACLE_ROCTX"))
+        # QUORACLE_CHECK=1: synchronize after every launch phase so async
+        # HIP faults surface at their source (SURVEY.md §5.2 debug mode)
+        self._strict_sync = bool(os.environ.get("QUORACLE_CHECK"))
 
     # -- lifecycle -----------------------------------------------------------
 
@@ -324,6 +327,8 @@ class LocalEngine(Engine):
                 self._crash_model(hm, exc)
                 continue
             if ctxt is not None:
+                if self._strict_sync and self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
                 pending.append((hm, ctxt))
         for hm, ctxt in pending:
             try:
