@@ -93,3 +93,26 @@ def cosine_sim_matrix(out, x):
 def gather_rows(out, src, rows):
     ext().gather_rows(out, src, rows)
     return out
+
+
+def paged_attn_decode_split(out, q, kcache, vcache, block_tables, ctx_lens,
+                            scale, part_m, part_l, part_acc):
+    ext().paged_attn_decode_split(out, q, kcache, vcache, block_tables,
+                                  ctx_lens, scale, part_m, part_l, part_acc)
+    return out
+
+
+def paged_attn_prefill_mfma(out, q, kcache, vcache, block_tables, tile_q0,
+                            tile_qn, tile_seq, tile_pos0, scale):
+    ext().paged_attn_prefill_mfma(out, q, kcache, vcache, block_tables,
+                                  tile_q0, tile_qn, tile_seq, tile_pos0,
+                                  scale)
+    return out
+
+
+def paged_attn_prefill_mfma32(out, q, kcache, vcache, block_tables, tile_q0,
+                              tile_qn, tile_seq, tile_pos0, scale):
+    ext().paged_attn_prefill_mfma32(out, q, kcache, vcache, block_tables,
+                                    tile_q0, tile_qn, tile_seq, tile_pos0,
+                                    scale)
+    return out
