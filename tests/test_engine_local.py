@@ -281,3 +281,15 @@ def test_anonymous_sessions_release_kv_on_finish():
                           timeout=120)
     assert r.ok
     assert eng.models["tiny#anon"].mgr.free_blocks == free0
+
+
+def test_embed_cache_ttl_expiry(monkeypatch):
+    from quoracle_amd.engine.pool import EmbedCache
+    t = {"now": 1000.0}
+    import quoracle_amd.engine.pool as pool_mod
+    monkeypatch.setattr(pool_mod._time, "monotonic", lambda: t["now"])
+    cache = EmbedCache(ttl_s=10.0, max_entries=10)
+    cache.put("x", [1.0])
+    assert cache.get("x") == [1.0]
+    t["now"] += 11.0
+    assert cache.get("x") is None          # expired
