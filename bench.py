@@ -75,6 +75,11 @@ def main():
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.gpus > 1 and world == 1:
+        print(f"[bench] WARNING: --gpus {args.gpus} but WORLD_SIZE=1 — "
+              f"multi-GPU runs must be launched via torchrun "
+              f"(--nproc-per-node {args.gpus}); continuing single-process",
+              file=sys.stderr, flush=True)
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
     if use_cuda:
