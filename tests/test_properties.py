@@ -133,3 +133,28 @@ def test_validator_todo_items_shape(items):
     for item in out["items"]:
         assert isinstance(item, dict)
         assert item.get("state") in (None, "todo", "pending", "done")
+
+
+from quoracle_amd.utils.jsonx import extract_json
+
+
+@given(st.text(max_size=200))
+@settings(max_examples=80, deadline=None)
+def test_extract_json_never_crashes(noise):
+    """Arbitrary junk around a valid object: extraction still finds it (or
+    returns None) without ever raising."""
+    assert extract_json(noise) is None or isinstance(extract_json(noise), dict)
+    embedded = noise + '{"action": "wait", "params": {}, "reasoning": "r"}'
+    out = extract_json(embedded)
+    # the balanced-span scan finds the object unless the noise itself
+    # contains an earlier '{' that breaks balance — never an exception
+    assert out is None or isinstance(out, dict)
+
+
+def test_extract_json_fenced_and_prefixed():
+    body = '{"action": "todo", "params": {"items": []}, "reasoning": "x"}'
+    assert extract_json(body)["action"] == "todo"
+    assert extract_json(f"Sure! Here:\n```json\n{body}\n```")["action"] == "todo"
+    assert extract_json(f"preamble {body} trailing")["action"] == "todo"
+    assert extract_json("no json here") is None
+    assert extract_json('{"broken": ') is None
