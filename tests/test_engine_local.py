@@ -293,3 +293,31 @@ def test_embed_cache_ttl_expiry(monkeypatch):
     assert cache.get("x") == [1.0]
     t["now"] += 11.0
     assert cache.get("x") is None          # expired
+
+
+def test_kv_exhaustion_under_pressure_fails_cleanly():
+    """With an artificially tiny KV pool, concurrent long generations either
+    evict idle sessions or fail with a structured kv_exhausted error — the
+    engine itself must survive and keep serving (engine.py _extend path)."""
+    eng = LocalEngine(["tiny"], device=torch.device("cpu"),
+                      embed_model_key=None, kv_blocks_override=24).start()
+    try:
+        reqs = [GenerateRequest(model_key="tiny",
+                                messages=[{"role": "user",
+                                           "content": "x" * 120}],
+                                max_tokens=48, temperature=0.0, seed=i,
+                                session_id=f"press-{i}")
+                for i in range(4)]
+        results = [eng.generate_sync(r, timeout=120) for r in reqs]
+        for r in results:
+            assert r.error is None or r.error.startswith(
+                ("kv_exhausted", "context_overflow")), r.error
+        # engine still alive and serving after the pressure burst
+        ok = eng.generate_sync(
+            GenerateRequest(model_key="tiny",
+                            messages=[{"role": "user", "content": "hi"}],
+                            max_tokens=4, temperature=0.0, seed=0),
+            timeout=120)
+        assert ok.error is None and ok.output_tokens >= 1
+    finally:
+        eng.stop()
