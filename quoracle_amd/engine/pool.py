@@ -94,6 +94,21 @@ class EnginePool:
     def models(self) -> List[str]:
         return list(self._by_model)
 
+    def engine_stats(self) -> Dict[str, int]:
+        """Aggregate telemetry over every distinct engine in the pool
+        (LocalEngine.stats counters summed; fakes/remotes without stats
+        contribute nothing). Feeds /metrics and /api/engine/stats."""
+        engines = []
+        for e in [self._default, self._embedder, *self._by_model.values()]:
+            if e is not None and e not in engines:
+                engines.append(e)
+        total: Dict[str, int] = {}
+        for e in engines:
+            for key, val in (getattr(e, "stats", None) or {}).items():
+                if isinstance(val, (int, float)):
+                    total[key] = total.get(key, 0) + val
+        return total
+
 
 class EmbedFacade:
     """Callable embed_many with an optional fused similarity_matrix —

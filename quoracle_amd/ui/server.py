@@ -221,8 +221,8 @@ def create_app(manager) -> FastAPI:
         emit("quoracle_tasks_total", len(tasks_))
         emit("quoracle_tasks_running",
              sum(1 for t in tasks_ if t.get("status") == "running"))
-        embedder = getattr(runtime.engines, "_embedder", None)
-        for key, val in (getattr(embedder, "stats", None) or {}).items():
+        stats_fn = getattr(runtime.engines, "engine_stats", None)
+        for key, val in (stats_fn() if stats_fn else {}).items():
             emit(f"quoracle_engine_{key}", val)
         total = runtime.store.total_cost(
             [a for t in tasks_ for a in
@@ -233,11 +233,8 @@ def create_app(manager) -> FastAPI:
 
     @app.get("/api/engine/stats")
     def engine_stats():
-        stats = {}
-        embedder = getattr(runtime.engines, "_embedder", None)
-        if embedder is not None and hasattr(embedder, "stats"):
-            stats = dict(embedder.stats)
-        return stats
+        stats_fn = getattr(runtime.engines, "engine_stats", None)
+        return stats_fn() if stats_fn else {}
 
     # -- event stream (PubSub tail) ------------------------------------------
 

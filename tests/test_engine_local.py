@@ -321,3 +321,21 @@ def test_kv_exhaustion_under_pressure_fails_cleanly():
         assert ok.error is None and ok.output_tokens >= 1
     finally:
         eng.stop()
+
+
+def test_engine_pool_stats_aggregates_distinct_engines():
+    from quoracle_amd.engine.pool import EnginePool
+
+    class _E:
+        def __init__(self, steps):
+            self.stats = {"engine_steps": steps, "requests_done": 1}
+
+        async def embed(self, texts):
+            return [[0.0]]
+
+    a, b = _E(10), _E(5)
+    pool = EnginePool(default=a, by_model={"m1": a, "m2": b})
+    stats = pool.engine_stats()
+    # a counted once despite appearing as default + m1
+    assert stats["engine_steps"] == 15
+    assert stats["requests_done"] == 2
