@@ -130,12 +130,17 @@ class ControlClient:
             rank, REQ_TAG, self._locks[rank])
         return await fut
 
-    def shutdown(self) -> None:
+    def shutdown(self, join_timeout: float = 30.0) -> None:
         for rank, lock in self._locks.items():
             try:
                 _send_obj({"kind": "stop"}, rank, REQ_TAG, lock)
             except Exception:
                 pass
+        # each server replies "bye" after engine.stop(); wait for the
+        # receiver threads to drain it — tearing down the process group
+        # while a thread is blocked in dist.recv aborts the process
+        for t in self._receivers:
+            t.join(timeout=join_timeout)
 
     def embed_gather(self, shards, counts) -> None:
         """Fan the per-rank text shards to the servers; every rank then
