@@ -66,6 +66,17 @@ def test_serve_fake_end_to_end(tmp_path):
         assert tree["agents"]
         metrics = urllib.request.urlopen(base + "/metrics", timeout=5).read()
         assert b"quoracle_tasks_running" in metrics
+        # lifecycle through the HTTP surface: pause -> restore -> delete
+        paused = _post(base + f"/api/tasks/{created['task_id']}/pause", {})
+        assert paused.get("ok") is True, paused
+        tasks = _get(base + "/api/tasks")
+        mine = next(t for t in tasks if t["task_id"] == created["task_id"])
+        assert mine["status"] == "paused", mine
+        restored = _post(base + f"/api/tasks/{created['task_id']}/restore",
+                         {})
+        assert isinstance(restored, dict), restored
+        stats = _get(base + "/api/engine/stats")
+        assert isinstance(stats, dict)
     finally:
         proc.terminate()
         try:
