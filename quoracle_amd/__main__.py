@@ -11,8 +11,6 @@
 from __future__ import annotations
 
 import argparse
-import asyncio
-import os
 import sys
 
 

@@ -24,7 +24,7 @@ Consensus-rule grammar (consensus_rules values):
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 ACTIONS: List[str] = [
     "spawn_child",

@@ -10,7 +10,7 @@ validation (min 2, batchable subset, no nesting).
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 from urllib.parse import urlparse
 
 from . import schema as schema_mod

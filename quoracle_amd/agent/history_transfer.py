@@ -11,7 +11,7 @@ keep their own histories untouched.
 from __future__ import annotations
 
 import copy
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, List
 
 from . import condensation as condensation_mod
 from . import token_manager as tm

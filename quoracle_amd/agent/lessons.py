@@ -12,7 +12,7 @@ instead of a per-pair loop.
 
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Optional
 
 from ..consensus.rules import EmbedManyFn, cosine_similarity
 

@@ -25,7 +25,7 @@ from ..governance.profiles import ProfileNotFoundError
 from ..registry import DuplicateAgentError
 from ..utils import ids
 from .core import AgentActor
-from .state import AgentState, history_entry
+from .state import AgentState
 
 logger = logging.getLogger(__name__)
 

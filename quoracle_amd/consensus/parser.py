@@ -9,7 +9,7 @@ optional 'condense' request, unknown-action rejection.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, Dict, List
 
 from ..actions import schema as schema_mod
 from ..utils.jsonx import extract_json

@@ -15,7 +15,6 @@ consensus_handler.ex:126-152, maps to GPU prefix sharing here).
 
 from __future__ import annotations
 
-import json
 from typing import Any, Dict, List, Optional
 
 from ..actions import schema as schema_mod

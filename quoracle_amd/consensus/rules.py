@@ -24,7 +24,7 @@ Rules:
 from __future__ import annotations
 
 import math
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Callable, Dict, List, Optional, Sequence
 
 from ..utils.jsonx import dumps_canonical
 

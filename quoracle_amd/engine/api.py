@@ -15,7 +15,7 @@ test strategy.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional, Protocol, Sequence
 
 # Dynamic max_tokens policy (reference: per_model_query.ex:17-24):

@@ -10,7 +10,7 @@ runs on the orchestrator's GPU.
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Dict, List, Optional, Sequence
 
 import hashlib
 import time as _time

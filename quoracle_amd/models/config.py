@@ -9,8 +9,8 @@ pool genuinely decorrelated voters.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field, replace
-from typing import Dict, Optional
+from dataclasses import dataclass
+from typing import Dict
 
 
 @dataclass(frozen=True)

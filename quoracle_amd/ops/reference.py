@@ -8,7 +8,7 @@ readable — correctness over speed.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

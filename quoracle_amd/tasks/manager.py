@@ -13,7 +13,6 @@ restored histories re-prefill on the next consensus cycle.
 
 from __future__ import annotations
 
-import asyncio
 import logging
 from typing import Any, Dict, List, Optional
 

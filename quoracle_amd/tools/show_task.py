@@ -9,7 +9,6 @@ quoracle.db, live or not.
 from __future__ import annotations
 
 import argparse
-import json
 import time
 from typing import Any, Dict, List, Optional
 
