@@ -6,8 +6,10 @@ turns them into multimodal history entries, compressing via libvips
 framework hosts text-only models (no vision pool member), so the MI355X
 rebuild DETECTS images (magic bytes / data URLs), stores them as artifacts
 on disk, and replaces the inline payload with a compact placeholder so huge
-binary blobs never enter a model history.  Compression is a documented
-divergence (no libvips in the image, no vision model to feed).
+binary blobs never enter a model history.  Oversized artifacts are
+compressed on the way to disk (Pillow: bounded max dimension, JPEG quality
+descent under a byte ceiling — the libvips-equivalent behavior); animated
+GIFs and undecodable payloads are stored verbatim.
 """
 
 from __future__ import annotations
@@ -31,6 +33,43 @@ _DATA_URL_RE = re.compile(
     r"data:(image/[a-z+.-]+);base64,([A-Za-z0-9+/=]{64,})")
 
 ARTIFACT_DIR_ENV = "QUORACLE_IMAGE_DIR"
+MAX_DIM = 2048               # longest side after compression
+MAX_BYTES = 5 * 1024 * 1024  # byte ceiling before compression kicks in
+
+
+def compress_image(data: bytes, mime: str,
+                   max_dim: int = MAX_DIM,
+                   max_bytes: int = MAX_BYTES) -> Tuple[bytes, str]:
+    """Bound an image's size (reference: utils/image_compressor.ex via
+    libvips; here Pillow): downscale so the longest side <= max_dim, then
+    JPEG-re-encode with descending quality until under max_bytes.  Returns
+    (data, mime) — unchanged when already small enough, undecodable, or
+    animated."""
+    if len(data) <= max_bytes:
+        return data, mime
+    try:
+        import io
+        from PIL import Image
+        img = Image.open(io.BytesIO(data))
+        if getattr(img, "is_animated", False):
+            return data, mime
+        img.load()
+    except Exception:  # noqa: BLE001 — store verbatim if not decodable
+        return data, mime
+    if max(img.size) > max_dim:
+        img.thumbnail((max_dim, max_dim))
+    if img.mode not in ("RGB", "L"):
+        img = img.convert("RGB")
+    best = data
+    for quality in (85, 70, 55, 40, 30):
+        buf = io.BytesIO()
+        img.save(buf, format="JPEG", quality=quality)
+        best = buf.getvalue()
+        if len(best) <= max_bytes:
+            break
+    if len(best) >= len(data):
+        return data, mime
+    return best, "image/jpeg"
 
 
 def sniff(data: bytes) -> Optional[str]:
@@ -43,6 +82,8 @@ def sniff(data: bytes) -> Optional[str]:
 
 
 def _store(data: bytes, mime: str) -> Dict[str, Any]:
+    original_bytes = len(data)
+    data, mime = compress_image(data, mime)
     digest = hashlib.sha256(data).hexdigest()[:16]
     ext = mime.split("/")[-1]
     out_dir = os.environ.get(ARTIFACT_DIR_ENV) or "/tmp/quoracle_images"
@@ -52,7 +93,7 @@ def _store(data: bytes, mime: str) -> Dict[str, Any]:
         with open(path, "wb") as f:
             f.write(data)
     return {"mime": mime, "bytes": len(data), "sha256_16": digest,
-            "path": path}
+            "path": path, "original_bytes": original_bytes}
 
 
 def extract_images(result: Any) -> Tuple[Any, List[Dict[str, Any]]]:

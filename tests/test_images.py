@@ -1,0 +1,83 @@
+"""Image artifact pipeline (utils/images.py; reference:
+agent/image_detector.ex + utils/image_compressor.ex): sniffing, data-URL
+extraction, and Pillow-based compression of oversized payloads."""
+
+import base64
+import io
+
+import pytest
+
+from quoracle_amd.utils import images as I
+
+
+def _png(w=32, h=32, color=(200, 30, 30)):
+    from PIL import Image
+    buf = io.BytesIO()
+    Image.new("RGB", (w, h), color).save(buf, format="PNG")
+    return buf.getvalue()
+
+
+def _noisy_png(w, h, seed=7):
+    import numpy as np
+    from PIL import Image
+    rng = np.random.default_rng(seed)
+    arr = rng.integers(0, 255, size=(h, w, 3), dtype="uint8")
+    buf = io.BytesIO()
+    Image.fromarray(arr, "RGB").save(buf, format="PNG")
+    return buf.getvalue()
+
+
+def test_sniff_magic_bytes():
+    assert I.sniff(_png()) == "image/png"
+    assert I.sniff(b"\xff\xd8\xff\xe0rest") == "image/jpeg"
+    assert I.sniff(b"GIF89a....") == "image/gif"
+    assert I.sniff(b"RIFFxxxxWEBP") == "image/webp"
+    assert I.sniff(b"RIFFxxxxWAVE") is None
+    assert I.sniff(b"plain text") is None
+
+
+def test_extract_data_url_and_bytes(tmp_path, monkeypatch):
+    monkeypatch.setenv(I.ARTIFACT_DIR_ENV, str(tmp_path))
+    png = _png()
+    url = "data:image/png;base64," + base64.b64encode(png).decode()
+    result = {"stdout": f"before {url} after", "raw": png, "n": 3}
+    cleaned, artifacts = I.extract_images(result)
+    assert len(artifacts) == 2
+    assert "image artifact" in cleaned["stdout"]
+    assert cleaned["stdout"].startswith("before ")
+    assert cleaned["stdout"].endswith(" after")
+    assert "image artifact" in cleaned["raw"]
+    assert cleaned["n"] == 3
+    for art in artifacts:
+        assert art["mime"] == "image/png"
+        with open(art["path"], "rb") as f:
+            assert f.read() == png
+
+
+def test_small_image_not_compressed():
+    png = _png()
+    data, mime = I.compress_image(png, "image/png")
+    assert data == png and mime == "image/png"
+
+
+def test_oversized_image_compressed(tmp_path, monkeypatch):
+    monkeypatch.setenv(I.ARTIFACT_DIR_ENV, str(tmp_path))
+    big = _noisy_png(3000, 3000)            # incompressible, > 5 MB
+    assert len(big) > I.MAX_BYTES
+    data, mime = I.compress_image(big, "image/png")
+    assert mime == "image/jpeg"
+    assert len(data) < len(big)
+    from PIL import Image
+    img = Image.open(io.BytesIO(data))
+    assert max(img.size) <= I.MAX_DIM       # downscaled
+    # through the artifact store: placeholder reflects compressed form
+    cleaned, arts = I.extract_images({"raw": big})
+    assert arts[0]["mime"] == "image/jpeg"
+    assert arts[0]["original_bytes"] == len(big)
+    assert arts[0]["bytes"] < len(big)
+
+
+def test_undecodable_oversized_payload_stored_verbatim():
+    blob = b"\x89PNG\r\n\x1a\n" + b"\x00" * (I.MAX_BYTES + 100)
+    data, mime = I.compress_image(blob, "image/png")
+    assert data == blob and mime == "image/png"
