@@ -70,3 +70,20 @@ def test_show_prompts_scenarios_render(name):
     from quoracle_amd.tools.show_prompts import render_scenario
     text = render_scenario(name)
     assert isinstance(text, str) and len(text) > 40
+
+
+def test_examples_demo_runs_clean():
+    """The examples walkthrough must stay executable (it doubles as living
+    documentation of the public API)."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "examples/demo.py"], cwd=REPO_ROOT,
+        capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "file_write completed" in out.stdout
+    assert "demo complete." in out.stdout
+
+
+import os as _os
+REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
