@@ -3,9 +3,15 @@ show-prompts tool (reference: lib/quoracle/agent/consensus_handler/*_injector.ex
 — blocks land in the last user message so the system-prompt KV prefix stays
 byte-stable across cycles)."""
 
+import os
+import subprocess
+import sys
+
 import pytest
 
 from quoracle_amd.agent import injectors as I
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_blocks_empty_inputs_render_empty():
@@ -75,8 +81,6 @@ def test_show_prompts_scenarios_render(name):
 def test_examples_demo_runs_clean():
     """The examples walkthrough must stay executable (it doubles as living
     documentation of the public API)."""
-    import subprocess
-    import sys
     out = subprocess.run(
         [sys.executable, "examples/demo.py"], cwd=REPO_ROOT,
         capture_output=True, text=True, timeout=180)
@@ -84,6 +88,3 @@ def test_examples_demo_runs_clean():
     assert "file_write completed" in out.stdout
     assert "demo complete." in out.stdout
 
-
-import os as _os
-REPO_ROOT = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
