@@ -158,3 +158,28 @@ def test_extract_json_fenced_and_prefixed():
     assert extract_json(f"preamble {body} trailing")["action"] == "todo"
     assert extract_json("no json here") is None
     assert extract_json('{"broken": ') is None
+
+
+# -- validator fuzz: no input shape may crash it ----------------------------
+
+_JSON_LEAF = st.one_of(st.none(), st.booleans(), st.integers(),
+                       st.floats(allow_nan=False), st.text(max_size=20))
+_JSON = st.recursive(
+    _JSON_LEAF,
+    lambda inner: st.one_of(st.lists(inner, max_size=4),
+                            st.dictionaries(st.text(max_size=8), inner,
+                                            max_size=4)),
+    max_leaves=10)
+
+
+@settings(max_examples=200, deadline=None)
+@given(params=_JSON, action_idx=st.integers(min_value=0, max_value=21))
+def test_validator_never_crashes_on_junk(params, action_idx):
+    from quoracle_amd.actions import schema as S
+    from quoracle_amd.actions.validator import ValidationError, validate_params
+    action = sorted(S.ACTIONS)[action_idx]
+    try:
+        out = validate_params(action, params)
+    except ValidationError:
+        return                     # rejection is a valid outcome
+    assert isinstance(out, dict)   # acceptance must yield coerced dict params
