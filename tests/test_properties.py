@@ -183,3 +183,29 @@ def test_validator_never_crashes_on_junk(params, action_idx):
     except ValidationError:
         return                     # rejection is a valid outcome
     assert isinstance(out, dict)   # acceptance must yield coerced dict params
+
+
+@settings(max_examples=60, deadline=None)
+@given(draws=st.lists(st.integers(min_value=0, max_value=2 ** 31 - 1),
+                      min_size=1, max_size=16),
+       subset=st.sets(st.integers(min_value=0, max_value=21), max_size=6),
+       ctx_depth=st.integers(min_value=0, max_value=3))
+def test_grammar_valid_over_any_action_subset(draws, subset, ctx_depth):
+    """The grammar invariant holds for EVERY subset of the 22 actions
+    (including the empty set -> wait fallback) and any context dict."""
+    from quoracle_amd.actions import schema as S
+    all_actions = sorted(S.ACTIONS)
+    allowed = [all_actions[i] for i in sorted(subset)]
+    context = {"task_description": "t" * ctx_depth,
+               "profile": "default"} if ctx_depth else None
+    g = ActionGrammar(allowed, reasoning_tokens=3, context=context)
+    out, i = [], 0
+    for _ in range(4000):
+        if g.done:
+            break
+        out.append(g.advance(draws[i % len(draws)]))
+        i += 1
+    assert g.done
+    parsed = json.loads(ByteTokenizer().decode([t for t in out if t != EOS]))
+    assert parsed["action"] in (allowed or ["wait"]) or \
+        parsed["action"] in g.candidates
