@@ -155,3 +155,35 @@ async def test_oversized_entry_recursively_summarized():
     big = [t for t in texts if "oversized entry summarized" in t]
     assert big
     assert eng.count_tokens(big[0]) < eng.count_tokens(giant)
+
+
+def test_ensure_fits_terminates_on_random_histories():
+    """Property: for arbitrary history shapes (including single giant
+    entries), ensure_fits terminates within its pass budget and only ever
+    shrinks the history (reference: per_model_query.ex:149-196)."""
+    import asyncio
+    import random
+    rng = random.Random(1234)
+    for trial in range(10):
+        n = rng.randint(1, 30)
+        history = [history_entry(rng.choice(["user_message", "action_result",
+                                             "event"]),
+                                 "x" * rng.randint(10, 30_000))
+                   for _ in range(n)]
+        state = AgentState(agent_id="a", task_id="t", profile="default",
+                           model_pool=["m"])
+        state.init_model_maps()
+        state.model_histories["m"] = list(history)
+        engine = FakeEngine(context_limits={"m": 6000})
+
+        def input_tokens():
+            return sum(engine.count_tokens(json.dumps(e, default=str))
+                       for e in state.model_histories["m"])
+
+        before = input_tokens()
+        out = asyncio.run(
+            cond.ensure_fits(state, "m", engine, input_tokens))
+        after = input_tokens()
+        assert after <= before
+        assert len(state.model_histories["m"]) >= 1
+        assert out == after
