@@ -189,3 +189,33 @@ async def test_show_task_transcript(tmp_path):
     assert "transcribe me" in text
     assert root_id in text
     assert "Agent tree" in text and "Messages" in text
+
+
+def test_store_concurrent_writes_from_threads(tmp_path):
+    """The store is shared across the asyncio runtime and engine threads;
+    hammer it from 8 threads and verify every row lands (WAL +
+    check_same_thread=False + lock)."""
+    import threading
+    from quoracle_amd.persistence.store import Store
+    store = Store(str(tmp_path / "conc.db"))
+    errors = []
+
+    def writer(tid):
+        try:
+            for i in range(50):
+                store.save_log(f"agent-{tid}", "task", "info", "evt",
+                               f"msg {i}")
+                store.save_cost(f"agent-{tid}", "task", "m", 0.01)
+        except Exception as exc:  # noqa: BLE001
+            errors.append(exc)
+
+    threads = [threading.Thread(target=writer, args=(t,)) for t in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errors, errors
+    for tid in range(8):
+        assert len(store.logs_for_agent(f"agent-{tid}", limit=100)) == 50
+        rows = store.costs_for_agent(f"agent-{tid}")
+        assert len(rows) == 50
