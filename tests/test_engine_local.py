@@ -339,3 +339,38 @@ def test_engine_pool_stats_aggregates_distinct_engines():
     # a counted once despite appearing as default + m1
     assert stats["engine_steps"] == 15
     assert stats["requests_done"] == 2
+
+
+def test_two_models_one_engine_interleave():
+    """Two hosted models on one engine: concurrent requests to both
+    complete, decode deterministically per (model, seed), and both models'
+    sessions survive (two-phase launch/sample per step)."""
+    eng = LocalEngine(["tiny", "gpt2s"], device=torch.device("cpu"),
+                      embed_model_key=None, kv_blocks_override=512).start()
+    try:
+        async def run():
+            reqs = []
+            for i in range(3):
+                for mk in ("tiny", "gpt2s"):
+                    reqs.append(GenerateRequest(
+                        model_key=mk,
+                        messages=[{"role": "user", "content": f"q{i}"}],
+                        max_tokens=6, temperature=0.8, seed=i,
+                        session_id=f"s{mk}{i}"))
+            return await asyncio.gather(*[eng.generate(r) for r in reqs])
+        results = asyncio.run(run())
+        assert all(r.ok for r in results), [r.error for r in results]
+        by_model = {}
+        for r in results:
+            by_model.setdefault(r.model_key, []).append(r)
+        assert set(by_model) == {"tiny", "gpt2s"}
+        # determinism: rerun one request -> same text
+        again = eng.generate_sync(GenerateRequest(
+            model_key="tiny", messages=[{"role": "user", "content": "q0"}],
+            max_tokens=6, temperature=0.8, seed=0, session_id="fresh0"),
+            timeout=120)
+        match = [r for r in results
+                 if r.model_key == "tiny"][0]
+        assert again.text == match.text
+    finally:
+        eng.stop()
