@@ -228,3 +228,36 @@ def test_every_action_schema_renders_and_validates_shape():
             assert p in sch.param_types, f"{name}.{p} missing type"
         for p in sch.required_params + sch.optional_params:
             assert isinstance(p, str) and p
+
+
+def test_system_prompt_section_order_and_stability():
+    """Prompt section order mirrors the reference assembly (identity ->
+    profile -> constraints -> schemas -> response format -> examples ->
+    skills; reference: consensus/prompt_builder.ex:91-134) and the output
+    is byte-stable for identical inputs (KV-prefix caching depends on it)."""
+    from quoracle_amd.consensus.prompt_builder import build_system_prompt
+    from quoracle_amd.governance.profiles import Profile
+    kwargs = dict(
+        role="researcher",
+        profile=Profile(name="p", description="test",
+                        model_pool=["m1", "m2"],
+                        capability_groups=["hierarchy"]),
+        constraints=["never delete files"],
+        capability_groups=["hierarchy", "file_read"],
+        skills=[{"name": "greet", "content": "Say hello."}],
+        agent_id="agent-1",
+    )
+    prompt = build_system_prompt(**kwargs)
+    anchors = ["autonomous agent", "researcher", "never delete files",
+               "spawn_child", "file_read", "reasoning", "greet"]
+    pos = [prompt.find(a) for a in anchors]
+    assert all(p >= 0 for p in pos), dict(zip(anchors, pos))
+    # identity before constraints before schemas before skills
+    assert pos[0] < pos[2] < pos[3] < pos[6]
+    # byte-stability across calls (prefix KV reuse)
+    assert prompt == build_system_prompt(**kwargs)
+    # forbidden actions are excluded from schemas
+    gated = build_system_prompt(capability_groups=["hierarchy"],
+                                forbidden_actions=["spawn_child"])
+    assert "### spawn_child" not in gated     # schema block removed
+    assert "### dismiss_child" in gated       # rest of the group intact
