@@ -67,8 +67,10 @@ def find_majority_cluster(
     return None
 
 
-def _batch_action_types(params: Dict[str, Any]) -> List[str]:
-    actions = params.get("actions")
+def _batch_action_types(params: Any) -> List[str]:
+    # total over malformed responses: clustering runs on parsed (not yet
+    # validated) model output, so params may be any JSON shape
+    actions = params.get("actions") if isinstance(params, dict) else None
     if not isinstance(actions, list):
         return []
     out = []
@@ -89,6 +91,8 @@ def action_fingerprint(response: Dict[str, Any]) -> Tuple[str, Any]:
     """
     action = response.get("action")
     params = response.get("params") or {}
+    if not isinstance(params, dict):
+        params = {}
     if action == "batch_async":
         return ("batch_async", sorted(_batch_action_types(params)))
     if action == "batch_sync":

@@ -209,3 +209,29 @@ def test_grammar_valid_over_any_action_subset(draws, subset, ctx_depth):
     parsed = json.loads(ByteTokenizer().decode([t for t in out if t != EOS]))
     assert parsed["action"] in (allowed or ["wait"]) or \
         parsed["action"] in g.candidates
+
+
+@settings(max_examples=120, deadline=None)
+@given(action_idx=st.integers(min_value=0, max_value=21), params=_JSON)
+def test_fingerprint_total_and_deterministic(action_idx, params):
+    """action_fingerprint never crashes on junk responses, is deterministic,
+    and is insensitive to unknown params (schema-normalized signature)."""
+    from quoracle_amd.actions import schema as S
+    from quoracle_amd.consensus.aggregator import action_fingerprint
+    action = sorted(S.ACTIONS)[action_idx]
+    resp = {"action": action, "params": params}
+    fp1 = action_fingerprint(resp)
+    fp2 = action_fingerprint(dict(resp))
+    assert fp1 == fp2
+    assert fp1[0] == action
+    if isinstance(params, dict):
+        noisy = dict(params)
+        noisy["__totally_unknown_param__"] = 123
+        assert action_fingerprint(
+            {"action": action, "params": noisy}) == fp1
+
+
+def test_fingerprint_unknown_action_is_invalid_cluster():
+    from quoracle_amd.consensus.aggregator import action_fingerprint
+    assert action_fingerprint({"action": "no_such", "params": {}}) == \
+        ("no_such", "invalid")
