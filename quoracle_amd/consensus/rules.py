@@ -108,6 +108,10 @@ def merge_wait_values(values: Sequence[Any]) -> Any:
         raise NoConsensus("no_values")
     booleans = [v for v in values if isinstance(v, bool)]
     integers = [v for v in values if isinstance(v, (int, float)) and not isinstance(v, bool)]
+    # drop unusable shapes ("wait": null / strings); all-unusable -> no wait
+    values = booleans + integers
+    if not values:
+        return False
 
     if not integers and booleans and all(b is False for b in booleans):
         return False
@@ -236,10 +240,13 @@ def _semantic_merge(
     return values[0]
 
 
-def _normalize_action_spec(spec: Dict[str, Any]) -> Dict[str, Any]:
+def _normalize_action_spec(spec: Any) -> Dict[str, Any]:
+    if not isinstance(spec, dict):
+        return {"action": None, "params": {}}
     action = spec.get("action")
-    params = spec.get("params") or {}
-    return {"action": action, "params": params}
+    params = spec.get("params")
+    return {"action": action,
+            "params": params if isinstance(params, dict) else {}}
 
 
 def _batch_sequence_merge(
@@ -253,6 +260,7 @@ def _batch_sequence_merge(
     """
     from ..actions import schema as schema_mod
 
+    sequences = [s for s in sequences if isinstance(s, list)]
     if not sequences:
         return []
     if len(sequences) == 1:

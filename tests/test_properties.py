@@ -235,3 +235,24 @@ def test_fingerprint_unknown_action_is_invalid_cluster():
     from quoracle_amd.consensus.aggregator import action_fingerprint
     assert action_fingerprint({"action": "no_such", "params": {}}) == \
         ("no_such", "invalid")
+
+
+_RULES = ["exact_match", ("semantic_similarity", 0.9), "mode_selection",
+          "union_merge", "structural_merge", ("percentile", 50),
+          "wait_parameter", "batch_sequence_merge"]
+
+
+@settings(max_examples=150, deadline=None)
+@given(rule_idx=st.integers(min_value=0, max_value=len(_RULES) - 1),
+       values=st.lists(_JSON, max_size=5))
+def test_merge_rules_total_over_junk(rule_idx, values):
+    """Every consensus rule either merges or raises NoConsensus on ANY
+    value shapes — a junk model response must never crash the merge."""
+    from quoracle_amd.consensus.rules import NoConsensus, apply_rule
+    from quoracle_amd.engine.fake import deterministic_embedding
+    def embed(texts):
+        return [deterministic_embedding(str(t)) for t in texts]
+    try:
+        apply_rule(_RULES[rule_idx], values, embed_many=embed)
+    except NoConsensus:
+        pass
