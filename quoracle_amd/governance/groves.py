@@ -100,7 +100,7 @@ def _rule_message(rule: Dict[str, Any]) -> str:
 def check_shell_command(command: str, hard_rules: Optional[List[Dict[str, Any]]],
                         skill_name: Optional[str] = None) -> None:
     for rule in hard_rules or []:
-        if rule.get("type") != "shell_pattern_block":
+        if not isinstance(rule, dict) or rule.get("type") != "shell_pattern_block":
             continue
         pattern = rule.get("pattern")
         if not isinstance(pattern, str) or not _rule_applies(rule, skill_name):
@@ -118,7 +118,7 @@ def check_shell_command(command: str, hard_rules: Optional[List[Dict[str, Any]]]
 def check_action(action: str, hard_rules: Optional[List[Dict[str, Any]]],
                  skill_name: Optional[str] = None) -> None:
     for rule in hard_rules or []:
-        if rule.get("type") != "action_block":
+        if not isinstance(rule, dict) or rule.get("type") != "action_block":
             continue
         actions = [a for a in rule.get("actions", []) if isinstance(a, str)]
         if _rule_applies(rule, skill_name) and action in actions:
@@ -267,7 +267,11 @@ def parse_grove_markdown(text: str, path: str = "") -> Dict[str, Any]:
         "governance": meta.get("governance"),
         "confinement": meta.get("confinement"),
         "confinement_mode": meta.get("confinement_mode"),
-        "hard_rules": meta.get("hard_rules") or [],
+        # non-dict rule entries are dropped at parse time so the enforcement
+        # hot path (check_shell_command/check_action) can trust the shape
+        "hard_rules": [r for r in (meta.get("hard_rules") or [])
+                       if isinstance(r, dict)]
+        if isinstance(meta.get("hard_rules"), list) else [],
         "schemas": meta.get("schemas"),
         "workspace": meta.get("workspace"),
         "skills_path": meta.get("skills_path"),

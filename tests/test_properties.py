@@ -256,3 +256,62 @@ def test_merge_rules_total_over_junk(rule_idx, values):
         apply_rule(_RULES[rule_idx], values, embed_many=embed)
     except NoConsensus:
         pass
+
+
+@settings(max_examples=60, deadline=None)
+@given(body=st.text(max_size=400))
+def test_grove_loader_total_over_junk_frontmatter(body, tmp_path_factory):
+    """load_grove on arbitrary GROVE.md content: parse or ValueError,
+    never a crash (governance input is user-editable)."""
+    import os
+    from quoracle_amd.governance.groves import load_grove
+    d = tmp_path_factory.mktemp("grove")
+    with open(os.path.join(str(d), "GROVE.md"), "w") as f:
+        f.write(body)
+    try:
+        grove = load_grove(str(d))
+        assert isinstance(grove, dict)
+    except (ValueError, KeyError):
+        pass
+
+
+@settings(max_examples=60, deadline=None)
+@given(rules=_JSON, confinement=_JSON)
+def test_grove_loader_total_over_junk_yaml_types(rules, confinement,
+                                                 tmp_path_factory):
+    """Structurally valid YAML frontmatter with arbitrarily WRONG types for
+    hard_rules/confinement must load-or-reject, never crash."""
+    import os
+    import yaml
+    from quoracle_amd.governance.groves import load_grove
+    d = tmp_path_factory.mktemp("grove")
+    front = yaml.safe_dump({"name": "g", "hard_rules": rules,
+                            "confinement": confinement},
+                           default_flow_style=True)
+    with open(os.path.join(str(d), "GROVE.md"), "w") as f:
+        f.write(f"---\n{front}---\nbody\n")
+    try:
+        grove = load_grove(str(d))
+        assert isinstance(grove, dict)
+    except (ValueError, KeyError):
+        pass
+
+
+@settings(max_examples=120, deadline=None)
+@given(hard_rules=_JSON, confinement=_JSON)
+def test_grove_enforcement_total_over_junk_rules(hard_rules, confinement):
+    """The gate-chain checks (shell/action/file/working-dir) must treat a
+    malformed grove as deny-or-allow, never crash at dispatch time."""
+    from quoracle_amd.governance import groves as G
+    rules = hard_rules if isinstance(hard_rules, list) else None
+    conf = confinement if isinstance(confinement, dict) else None
+    for fn, args in [
+        (G.check_shell_command, ("rm -rf /", rules)),
+        (G.check_action, ("spawn_child", rules)),
+        (G.check_shell_working_dir, ("/tmp/x", conf)),
+        (G.check_file_access, ("/tmp/y.txt", "write", conf)),
+    ]:
+        try:
+            fn(*args)
+        except (G.HardRuleViolation, G.ConfinementViolation):
+            pass
