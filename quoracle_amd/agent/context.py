@@ -24,7 +24,14 @@ ROLE_BY_TYPE = {
 
 
 def _ts_str(ts: Optional[float]) -> str:
-    return time.strftime("%Y-%m-%d %H:%M:%S UTC", time.gmtime(ts or time.time()))
+    if not isinstance(ts, (int, float)):
+        ts = None
+    try:
+        return time.strftime("%Y-%m-%d %H:%M:%S UTC",
+                             time.gmtime(ts or time.time()))
+    except (OSError, OverflowError, ValueError):
+        # out-of-range timestamp from a corrupted checkpoint
+        return time.strftime("%Y-%m-%d %H:%M:%S UTC", time.gmtime())
 
 
 def _format_content(entry: Dict[str, Any]) -> str:

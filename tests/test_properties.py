@@ -315,3 +315,45 @@ def test_grove_enforcement_total_over_junk_rules(hard_rules, confinement):
             fn(*args)
         except (G.HardRuleViolation, G.ConfinementViolation):
             pass
+
+
+_ENTRY = st.one_of(
+    _JSON,
+    st.fixed_dictionaries({"type": st.text(max_size=10)},
+                          optional={"content": _JSON,
+                                    "ts": st.one_of(st.none(), st.floats(
+                                        allow_nan=False))}))
+
+
+@settings(max_examples=100, deadline=None)
+@given(history=st.lists(_ENTRY, max_size=8))
+def test_context_build_total_over_corrupt_history(history):
+    """A corrupted checkpoint (arbitrary JSON in model_histories) must not
+    crash message building — the restore path renders whatever it finds."""
+    from quoracle_amd.agent.context import (build_conversation_messages,
+                                            merge_consecutive)
+    entries = [e for e in history if isinstance(e, dict)]
+    msgs = build_conversation_messages(entries)
+    assert isinstance(msgs, list)
+    for m in msgs:
+        assert m["role"] in ("system", "user", "assistant")
+        assert isinstance(m["content"], str)
+    merged = merge_consecutive(msgs)
+    # alternation guard: no two consecutive same-role messages
+    for a, b in zip(merged, merged[1:]):
+        assert not (a["role"] == b["role"] and a["role"] != "system")
+
+
+@settings(max_examples=100, deadline=None)
+@given(history=st.lists(_ENTRY, max_size=8))
+def test_token_manager_total_over_corrupt_history(history):
+    from quoracle_amd.agent.token_manager import (entry_tokens,
+                                                  history_tokens,
+                                                  split_for_condensation)
+    from quoracle_amd.engine.tokenizer import ByteTokenizer
+    count = ByteTokenizer().count
+    entries = [e for e in history if isinstance(e, dict)]
+    total = history_tokens(count, entries)
+    assert total >= 0
+    keep, discarded = split_for_condensation(count, entries)
+    assert len(keep) + len(discarded) == len(entries)
