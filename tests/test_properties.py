@@ -357,3 +357,21 @@ def test_token_manager_total_over_corrupt_history(history):
     assert total >= 0
     keep, discarded = split_for_condensation(count, entries)
     assert len(keep) + len(discarded) == len(entries)
+
+
+@settings(max_examples=100, deadline=None)
+@given(extra=st.dictionaries(st.text(max_size=12), _JSON, max_size=6))
+def test_checkpoint_roundtrip_tolerates_extra_junk(extra):
+    """from_checkpoint over a valid checkpoint polluted with junk keys
+    (schema drift between versions) restores the known fields and ignores
+    the rest."""
+    from quoracle_amd.agent.state import AgentState
+    state = AgentState(agent_id="a1", task_id="t1", profile="default",
+                       model_pool=["m"])
+    state.init_model_maps()
+    ckpt = state.to_checkpoint()
+    polluted = {**extra, **ckpt}
+    restored = AgentState.from_checkpoint(polluted)
+    assert restored.agent_id == "a1" and restored.task_id == "t1"
+    assert restored.model_pool == ["m"]
+    assert restored.to_checkpoint() == ckpt
