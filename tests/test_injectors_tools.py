@@ -88,3 +88,27 @@ def test_examples_demo_runs_clean():
     assert "file_write completed" in out.stdout
     assert "demo complete." in out.stdout
 
+
+
+def test_show_task_sparse_db(tmp_path):
+    """render_task on a task with no agents/logs degrades gracefully."""
+    from quoracle_amd.persistence.store import Store
+    from quoracle_amd.tools.show_task import render_task
+    store = Store(str(tmp_path / "sparse.db"))
+    store.save_task({"task_id": "t-empty", "status": "created",
+                     "prompt": "bare prompt", "profile": "default"})
+    text = render_task(store, "t-empty")
+    assert "bare prompt" in text
+    missing = render_task(store, "t-nope")
+    assert isinstance(missing, str)
+
+
+def test_event_history_ring_buffer_bounded():
+    from quoracle_amd.events import EventBus, LOG_HISTORY_LIMIT
+    bus = EventBus()
+    for i in range(LOG_HISTORY_LIMIT + 50):
+        bus.broadcast("agents:a:logs", "log", {"i": i})
+    hist = bus.history("agents:a:logs")
+    assert len(hist) == LOG_HISTORY_LIMIT
+    assert hist[-1].payload["i"] == LOG_HISTORY_LIMIT + 49
+    assert bus.history("agents:a:unknown") == []
