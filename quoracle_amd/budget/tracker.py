@@ -9,6 +9,7 @@ decreases must not go below what the child has already spent+committed.
 
 from __future__ import annotations
 
+import math
 from dataclasses import dataclass
 from typing import Optional
 
@@ -52,7 +53,8 @@ def parse_amount(value) -> float:
         amount = float(value)
     except (TypeError, ValueError):
         raise BudgetError("invalid_amount") from None
-    if amount <= 0:
+    # NaN/inf would poison every downstream escrow comparison
+    if not math.isfinite(amount) or amount <= 0:
         raise BudgetError("invalid_amount")
     return amount
 

@@ -375,3 +375,43 @@ def test_checkpoint_roundtrip_tolerates_extra_junk(extra):
     assert restored.agent_id == "a1" and restored.task_id == "t1"
     assert restored.model_pool == ["m"]
     assert restored.to_checkpoint() == ckpt
+
+
+@settings(max_examples=200, deadline=None)
+@given(allocated=st.floats(min_value=0.01, max_value=1e6),
+       ops=st.lists(st.tuples(st.sampled_from(["spend", "lock", "release"]),
+                               st.floats(min_value=0.001, max_value=1e6)),
+                    max_size=20))
+def test_budget_escrow_invariant(allocated, ops):
+    """Property: through any sequence of spend/escrow/release operations the
+    tracker never lets available go negative without raising, and committed
+    never underflows below zero."""
+    from quoracle_amd.budget.tracker import (BudgetError, BudgetView,
+                                             check_can_spend,
+                                             lock_allocation,
+                                             release_allocation)
+    view = BudgetView(mode="allocated", allocated=allocated,
+                      spent=0.0, committed=0.0)
+    for op, amount in ops:
+        try:
+            if op == "spend":
+                check_can_spend(view, amount)
+                view.spent += amount
+            elif op == "lock":
+                view.committed = lock_allocation(view, amount)
+            else:
+                view.committed = release_allocation(view.committed, amount)
+        except BudgetError:
+            continue
+        assert view.committed >= 0
+        assert view.available >= -1e-6, (op, amount, view)
+
+
+def test_parse_amount_rejects_junk():
+    from quoracle_amd.budget.tracker import BudgetError, parse_amount
+    import pytest as _pytest
+    assert parse_amount("50.00") == 50.0
+    for bad in ("nope", None, -1, 0, [], {}, "nan", "inf", "-inf",
+                float("nan"), float("inf")):
+        with _pytest.raises(BudgetError):
+            parse_amount(bad)
