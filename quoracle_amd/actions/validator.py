@@ -10,6 +10,8 @@ validation (min 2, batchable subset, no nesting).
 
 from __future__ import annotations
 
+import math
+
 from typing import Any, Dict, List, Optional
 from urllib.parse import urlparse
 
@@ -160,6 +162,10 @@ def _validate_type(value: Any, expected: Any) -> Any:
         raise ValidationError("invalid_param_type")
     if expected == "number":
         if isinstance(value, (int, float)) and not isinstance(value, bool):
+            # Python's json accepts NaN/Infinity literals; they would poison
+            # merges (median/percentile) and timers downstream
+            if isinstance(value, float) and not math.isfinite(value):
+                raise ValidationError("invalid_param_type")
             return value
         raise ValidationError("invalid_param_type")
     if expected == "boolean":

@@ -107,8 +107,10 @@ def merge_wait_values(values: Sequence[Any]) -> Any:
     if not values:
         raise NoConsensus("no_values")
     booleans = [v for v in values if isinstance(v, bool)]
-    integers = [v for v in values if isinstance(v, (int, float)) and not isinstance(v, bool)]
-    # drop unusable shapes ("wait": null / strings); all-unusable -> no wait
+    import math as _math
+    integers = [v for v in values if isinstance(v, (int, float))
+                and not isinstance(v, bool) and _math.isfinite(v)]
+    # drop unusable shapes ("wait": null / NaN / strings); all-unusable -> no wait
     values = booleans + integers
     if not values:
         return False

@@ -164,3 +164,20 @@ class TestTemperature:
 
     def test_invalid_round(self):
         assert round_temperature("llama", 0) == 1.0
+
+
+def test_nan_and_infinity_rejected_in_number_params():
+    """Python json parses NaN/Infinity literals; the validator must reject
+    them before they poison merges or timers."""
+    import pytest as _pytest
+    from quoracle_amd.actions.validator import ValidationError, validate_params
+    for bad in (float("nan"), float("inf"), float("-inf")):
+        with _pytest.raises(ValidationError):
+            validate_params("wait", {"wait": bad})
+    assert validate_params("wait", {"wait": 5.0}) == {"wait": 5.0}
+
+
+def test_wait_merge_filters_nonfinite():
+    from quoracle_amd.consensus.rules import merge_wait_values
+    assert merge_wait_values([float("nan"), 10, 20]) == 15
+    assert merge_wait_values([float("nan"), float("inf")]) is False
