@@ -415,3 +415,17 @@ def test_parse_amount_rejects_junk():
                 float("nan"), float("inf")):
         with _pytest.raises(BudgetError):
             parse_amount(bad)
+
+
+@settings(max_examples=150, deadline=None)
+@given(text=st.one_of(st.text(max_size=300),
+                      _JSON.map(lambda v: __import__("json").dumps(v))))
+def test_parse_response_total(text):
+    """parse_response over arbitrary model text: a parsed dict or
+    ParseError, never a crash."""
+    from quoracle_amd.consensus.parser import ParseError, parse_response
+    try:
+        out = parse_response(text)
+        assert isinstance(out, dict) and "action" in out
+    except ParseError:
+        pass
