@@ -68,3 +68,53 @@ def test_control_plane_generate_roundtrip():
         p.join(timeout=120)
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
     assert ok, errors
+
+
+def _worker3(rank, world, port, out_q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from quoracle_amd.engine.api import GenerateRequest
+        from quoracle_amd.engine.engine import LocalEngine
+        from quoracle_amd.parallel.control import (ControlClient,
+                                                   RemoteEngine, serve_engine)
+        if rank != 0:
+            eng = LocalEngine(["tiny"], device=torch.device("cpu"),
+                              embed_model_key=None)
+            serve_engine(eng)
+            return
+        client = ControlClient(remote_ranks=[1, 2])
+        remotes = {r: RemoteEngine(r, client) for r in (1, 2)}
+
+        async def drive():
+            reqs = [(r, GenerateRequest(
+                model_key="tiny",
+                messages=[{"role": "user", "content": f"m{r}-{i}"}],
+                max_tokens=3, temperature=0.0, seed=i))
+                for r in (1, 2) for i in range(3)]
+            return await asyncio.gather(
+                *[remotes[r].generate(q) for r, q in reqs])
+
+        results = asyncio.run(drive())
+        client.shutdown()
+        out_q.put((all(r.error is None for r in results), len(results)))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_control_plane_two_servers():
+    """World 3: rank 0 fans out to two independent engine servers with
+    concurrent in-flight requests (the shape of the driver's 8-GPU run)."""
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker3, args=(r, 3, 29554, out_q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    ok, n = out_q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=120)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    assert ok and n == 6
