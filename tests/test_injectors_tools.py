@@ -112,3 +112,15 @@ def test_event_history_ring_buffer_bounded():
     assert len(hist) == LOG_HISTORY_LIMIT
     assert hist[-1].payload["i"] == LOG_HISTORY_LIMIT + 49
     assert bus.history("agents:a:unknown") == []
+
+
+def test_examples_demo_governance_runs_clean():
+    out = subprocess.run(
+        [sys.executable, "examples/demo_governance.py"], cwd=REPO_ROOT,
+        capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "hard_rule_violation" in out.stdout
+    assert "confinement_violation" in out.stdout
+    assert "confined write landed: True" in out.stdout
+    assert "REDACTED marker present: True" in out.stdout
+    assert "leaked into logs/history: False" in out.stdout
