@@ -124,7 +124,9 @@ class AgentState:
         state = cls(agent_id=data["agent_id"], task_id=data["task_id"],
                     parent_id=data.get("parent_id"))
         for key, value in data.items():
-            if hasattr(state, key):
+            # only public data fields: a corrupt/malicious checkpoint must
+            # not reach __dict__/__class__ or private attributes
+            if not key.startswith("_") and hasattr(state, key):
                 setattr(state, key, value)
         state.init_model_maps()
         return state
