@@ -374,3 +374,29 @@ def test_two_models_one_engine_interleave():
         assert again.text == match.text
     finally:
         eng.stop()
+
+
+def test_engine_pool_unknown_model_raises():
+    from quoracle_amd.engine.pool import EnginePool
+    import pytest as _pytest
+    pool = EnginePool()
+    with _pytest.raises(KeyError):
+        pool.engine_for("ghost-model")
+
+
+def test_event_bus_unsubscribe_idempotent():
+    from quoracle_amd.events import EventBus
+    bus = EventBus()
+    q = bus.subscribe("t")
+    bus.unsubscribe("t", q)
+    bus.unsubscribe("t", q)          # second call is a no-op
+    bus.broadcast("t", "x", {})      # no deliveries, no crash
+    assert q.empty()
+
+
+def test_id_generators_unique_and_prefixed():
+    from quoracle_amd.utils import ids
+    agent_ids = {ids.agent_id("root") for _ in range(500)}
+    assert len(agent_ids) == 500
+    assert all(i.startswith("root_") for i in agent_ids)
+    assert ids.agent_id("agent") != ids.agent_id("agent")
