@@ -429,3 +429,16 @@ def test_parse_response_total(text):
         assert isinstance(out, dict) and "action" in out
     except ParseError:
         pass
+
+
+@settings(max_examples=80, deadline=None)
+@given(text=st.text(max_size=400))
+def test_skill_parser_total(text):
+    """SKILL.md content is user-authored (and agent-authored via
+    create_skill): parse-or-reject, never crash."""
+    from quoracle_amd.governance.skills import parse_skill_markdown
+    try:
+        out = parse_skill_markdown(text)
+        assert isinstance(out, dict)
+    except (ValueError, KeyError):
+        pass
