@@ -436,9 +436,9 @@ def test_parse_response_total(text):
 def test_skill_parser_total(text):
     """SKILL.md content is user-authored (and agent-authored via
     create_skill): parse-or-reject, never crash."""
-    from quoracle_amd.governance.skills import parse_skill_markdown
+    from quoracle_amd.governance.skills import SkillError, parse_skill_markdown
     try:
         out = parse_skill_markdown(text)
         assert isinstance(out, dict)
-    except (ValueError, KeyError):
+    except (SkillError, ValueError, KeyError):
         pass
