@@ -439,6 +439,10 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
             return _err("unknown_connection_id")
         if params.get("terminate"):
             conn.proc.terminate()
+            try:
+                await asyncio.wait_for(conn.proc.wait(), 5)
+            except (asyncio.TimeoutError, ProcessLookupError):
+                conn.proc.kill()
             agent.mcp_connections.pop(params["connection_id"], None)
             return {"connection_id": conn.connection_id, "status": "terminated"}
         tool = params.get("tool")
@@ -476,6 +480,10 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
             conn.tools = (tools_result or {}).get("tools", [])
         except Exception as exc:  # noqa: BLE001
             proc.terminate()
+            try:
+                await asyncio.wait_for(proc.wait(), 5)
+            except (asyncio.TimeoutError, ProcessLookupError):
+                proc.kill()
             return _err("mcp_connect_failed", detail=str(exc))
         agent.mcp_connections[conn.connection_id] = conn
         return {"connection_id": conn.connection_id,

@@ -616,6 +616,7 @@ class AgentActor:
         for conn in list(self.mcp_connections.values()):
             try:
                 conn.proc.terminate()
+                await asyncio.wait_for(conn.proc.wait(), 5)
             except Exception:
                 pass
         # release the engine-side prefix-cache sessions (KV blocks) this
