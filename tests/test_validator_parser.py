@@ -181,3 +181,13 @@ def test_wait_merge_filters_nonfinite():
     from quoracle_amd.consensus.rules import merge_wait_values
     assert merge_wait_values([float("nan"), 10, 20]) == 15
     assert merge_wait_values([float("nan"), float("inf")]) is False
+
+
+def test_tokenizer_edges():
+    from quoracle_amd.engine.tokenizer import EOS, ByteTokenizer
+    tok = ByteTokenizer()
+    assert tok.count("") == 0
+    assert tok.encode("") == []
+    assert tok.decode([]) == ""
+    assert tok.decode([EOS]) == ""          # EOS renders empty
+    assert tok.count("héllo") == len("héllo".encode())
