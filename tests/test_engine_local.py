@@ -400,3 +400,21 @@ def test_id_generators_unique_and_prefixed():
     assert len(agent_ids) == 500
     assert all(i.startswith("root_") for i in agent_ids)
     assert ids.agent_id("agent") != ids.agent_id("agent")
+
+
+def test_dynamic_max_tokens_policy():
+    """context − 1.12×input, floor MIN_OUTPUT_TOKENS, cap output limit
+    (reference: per_model_query.ex:136-145)."""
+    from quoracle_amd.engine.api import (MIN_OUTPUT_TOKENS,
+                                         TOKEN_SAFETY_MARGIN,
+                                         dynamic_max_tokens)
+    from quoracle_amd.engine.fake import FakeEngine
+    eng = FakeEngine(context_limits={"m": 40_000},
+                     output_limits={"m": 8_000})
+    # small input: capped by output limit
+    assert dynamic_max_tokens(eng, "m", 100) == 8_000
+    # large input: linear budget
+    assert dynamic_max_tokens(eng, "m", 30_000) == \
+        40_000 - int(30_000 * TOKEN_SAFETY_MARGIN)
+    # overflow-sized input: floored at the minimum, never negative
+    assert dynamic_max_tokens(eng, "m", 200_000) == MIN_OUTPUT_TOKENS
