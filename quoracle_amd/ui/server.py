@@ -279,7 +279,8 @@ h2{font-size:13px;color:#7af;margin:4px 0}
 textarea,input,select{width:95%;background:#222;color:#ddd;border:1px solid #444}
 button{background:#247;color:#fff;border:0;padding:4px 10px;cursor:pointer}
 </style></head><body>
-<div class=panel id=left><h2>tasks</h2><div id=tasks></div>
+<div class=panel id=left><h2>engine</h2><div id=estats style="font-size:11px;color:#9c9"></div>
+<h2>tasks</h2><div id=tasks></div>
 <h2>new task</h2><textarea id=prompt rows=3></textarea>
 <input id=profile value=default placeholder=profile>
 <button onclick=createTask()>create</button></div>
@@ -291,6 +292,10 @@ button{background:#247;color:#fff;border:0;padding:4px 10px;cursor:pointer}
 let selTask=null, selAgent=null;
 async function j(u,opt){const r=await fetch(u,opt);return r.json()}
 async function refresh(){
+ try{const es=await j('/api/engine/stats');
+  document.getElementById('estats').textContent=
+   Object.entries(es).map(([k,v])=>`${k}=${v}`).join('  ')||'no engine stats';
+ }catch(e){}
  const tasks=await j('/api/tasks');
  let html='';
  for(const t of tasks){
