@@ -113,6 +113,26 @@ def create_app(manager) -> FastAPI:
                           "alive": entry is not None})
         return {"task_id": task_id, "agents": nodes}
 
+    @app.get("/api/tasks/{task_id}/export")
+    def task_export(task_id: str):
+        """Full machine-readable transcript: task row, agent rows (with
+        checkpointed state), logs, messages, cost rollup — the JSON
+        counterpart of the show-task CLI renderer."""
+        task = runtime.store.get_task(task_id)
+        if task is None:
+            raise HTTPException(404, "unknown task")
+        agents = runtime.store.agents_for_task(task_id)
+        return {
+            "task": task,
+            "agents": [{
+                **a,
+                "logs": runtime.store.logs_for_agent(a["agent_id"],
+                                                     limit=200),
+                "costs": runtime.store.cost_rollup(a["agent_id"]),
+            } for a in agents],
+            "messages": runtime.store.messages_for_task(task_id),
+        }
+
     @app.get("/api/tasks/{task_id}/messages")
     def task_messages(task_id: str):
         return runtime.store.messages_for_task(task_id)

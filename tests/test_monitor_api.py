@@ -137,3 +137,23 @@ def test_full_task_work_fields(client, tmp_path):
                     for e in h)
         time.sleep(0.05)
     assert found
+
+
+def test_task_export_endpoint(client):
+    c, runtime = client
+    created = c.post("/api/tasks", json={"prompt": "export me",
+                                         "profile": "default"}).json()
+    task_id = created["task_id"]
+    import time
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        out = c.get(f"/api/tasks/{task_id}/export").json()
+        if out["agents"] and any(a["logs"] for a in out["agents"]):
+            break
+        time.sleep(0.1)
+    assert out["task"]["task_id"] == task_id
+    assert out["task"]["prompt"] == "export me"
+    assert out["agents"][0]["agent_id"] == created["root_agent_id"]
+    assert isinstance(out["agents"][0]["costs"], (dict, list))
+    assert isinstance(out["messages"], list)
+    assert c.get("/api/tasks/t-ghost/export").status_code == 404
