@@ -311,3 +311,24 @@ def test_vault_unseal_total_over_corrupt_blobs():
         raise AssertionError("wrong key must not decrypt")
     except SecretNotFoundError:
         pass
+
+
+def test_profile_catalog_shape():
+    from quoracle_amd.governance.profiles import Profile, ProfileStore
+    store = ProfileStore()
+    store.put(Profile(name="researcher", description="digs deep",
+                      model_pool=["m"], capability_groups=[]))
+    cat = store.catalog()
+    assert {"name": "researcher", "description": "digs deep"} in cat
+    assert all(set(c) == {"name", "description"} for c in cat)
+
+
+def test_registry_duplicate_id_raises():
+    import pytest as _pytest
+    from quoracle_amd.registry import Registry
+    reg = Registry()
+    reg.register("dup", object(), "t1")
+    with _pytest.raises(Exception):
+        reg.register("dup", object(), "t1")
+    reg.unregister("dup")
+    reg.register("dup", object(), "t1")   # reusable after unregister
