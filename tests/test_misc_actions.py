@@ -219,3 +219,27 @@ def test_store_concurrent_writes_from_threads(tmp_path):
         assert len(store.logs_for_agent(f"agent-{tid}", limit=100)) == 50
         rows = store.costs_for_agent(f"agent-{tid}")
         assert len(rows) == 50
+
+
+def test_cost_rollup_recursive_cte(tmp_path):
+    """Descendant costs aggregate through the recursive CTE over parent
+    links (reference: costs/aggregator.ex:122-163): root sees the whole
+    subtree, a middle node sees only its own branch."""
+    from quoracle_amd.persistence.store import Store
+    store = Store(str(tmp_path / "cte.db"))
+    #   root -> a -> a1
+    #        -> b
+    store.save_agent("root", "t", None, config={}, state={}, status="running")
+    store.save_agent("a", "t", "root", config={}, state={}, status="running")
+    store.save_agent("a1", "t", "a", config={}, state={}, status="running")
+    store.save_agent("b", "t", "root", config={}, state={}, status="running")
+    store.save_cost("root", "t", "m1", 1.0)
+    store.save_cost("a", "t", "m1", 2.0)
+    store.save_cost("a1", "t", "m2", 4.0)
+    store.save_cost("b", "t", "m1", 8.0)
+    root_roll = store.cost_rollup("root")
+    assert abs(root_roll["total"] - 15.0) < 1e-9
+    a_roll = store.cost_rollup("a")
+    assert abs(a_roll["total"] - 6.0) < 1e-9
+    leaf = store.cost_rollup("a1")
+    assert abs(leaf["total"] - 4.0) < 1e-9
