@@ -196,7 +196,8 @@ def apply_rule(
 
     if isinstance(rule, tuple) and rule[0] == "percentile":
         numeric = [v for v in values
-                   if isinstance(v, (int, float)) and not isinstance(v, bool)]
+                   if isinstance(v, (int, float)) and not isinstance(v, bool)
+                   and math.isfinite(v)]
         if not numeric:
             return mode_value(values)
         return _percentile(sorted(numeric), rule[1])
