@@ -128,8 +128,12 @@ async def shrink_oversized_entries(state: AgentState, model_key: str,
         if engine.count_tokens(content) <= budget:
             continue
         summary = await summarize_text(engine, model_key, content, budget)
-        history[i] = {**entry, "content":
-                      "[oversized entry summarized]\n" + summary}
+        new_entry = {**entry, "content":
+                     "[oversized entry summarized]\n" + summary}
+        # content changed: the memoized token count (token_manager.entry_tokens)
+        # would otherwise carry the stale oversized figure forever
+        new_entry.pop("_tokens", None)
+        history[i] = new_entry
         changed = True
     return changed
 

@@ -100,11 +100,15 @@ class Supervisor:
     async def spawn_child_action(self, parent: AgentActor,
                                  params: Dict[str, Any]) -> Dict[str, Any]:
         state = parent.state
-        child_id = ids.agent_id()
 
-        # dismiss-vs-spawn race guard (reference: spawn.ex:76-97)
-        if child_id in state.dismissing:
-            return {"error": "dismissing"}
+        # dismiss-vs-spawn race guard: refuse to spawn while the parent has
+        # ANY dismissal in flight (reference: spawn.ex:73-107 returns
+        # :parent_dismissing whenever the parent is dismissing children)
+        if state.dismissing:
+            return {"error": "parent_dismissing",
+                    "dismissing": sorted(state.dismissing)}
+
+        child_id = ids.agent_id()
 
         # Topology contract: grove may auto-inject profile/skills/constraints
         parent_skill_names = [s.get("name") for s in state.active_skills]
@@ -207,7 +211,7 @@ class Supervisor:
         if len(raw) <= 1200:
             return raw
         model = (self.runtime.config.model_roles.get("summarization")
-                 or state.model_pool[0] if state.model_pool else None)
+                 or (state.model_pool[0] if state.model_pool else None))
         if model is None:
             return raw[:1200]
         try:

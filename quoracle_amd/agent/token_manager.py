@@ -28,16 +28,19 @@ def entry_text(entry: Dict[str, Any]) -> str:
 
 
 def entry_tokens(count_tokens, entry: Dict[str, Any]) -> int:
-    """Token count of one history entry, memoized on the entry (entries are
-    append-only, so the count never goes stale; the byte tokenizer makes it
-    deterministic).  Keeps history_tokens O(new entries) per cycle instead
-    of O(history)."""
+    """Token count of one history entry, memoized on the entry.  The memo is
+    keyed by content length so any in-place rewrite (e.g. condensation's
+    shrink_oversized_entries) invalidates it instead of over-reporting
+    forever.  Keeps history_tokens O(new entries) per cycle instead of
+    O(history)."""
+    text = entry_text(entry)
     cached = entry.get("_tokens")
-    if isinstance(cached, int):
+    if isinstance(cached, int) and entry.get("_tokens_len") == len(text):
         return cached
-    n = count_tokens(entry_text(entry))
+    n = count_tokens(text)
     try:
         entry["_tokens"] = n
+        entry["_tokens_len"] = len(text)
     except TypeError:
         pass
     return n

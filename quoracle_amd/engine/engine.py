@@ -274,8 +274,10 @@ class LocalEngine(Engine):
                          hm.cfg.max_context - len(prompt))
         if request.action_grammar:
             # the grammar terminates by itself (~600 tokens worst case);
-            # never truncate it into unparseable JSON
-            max_tokens = max(max_tokens, 1024)
+            # never truncate it into unparseable JSON — but stay inside the
+            # model's remaining window or positions run past max_context
+            max_tokens = min(max(max_tokens, 1024),
+                             hm.cfg.max_context - len(prompt))
         params = SamplingParams(
             temperature=request.temperature, top_p=request.top_p,
             max_tokens=max_tokens, seed=request.seed)

@@ -125,6 +125,14 @@ class EmbedFacade:
         # cache hits are free, only fresh embeddings count
         self.embedded_tokens = 0
 
+    def _count(self, text: str) -> int:
+        """Exact token count via the engine's own tokenizer (the byte
+        tokenizer is one call away — never estimate with len//4)."""
+        try:
+            return max(1, self._engine.count_tokens(text))
+        except Exception:  # noqa: BLE001 — engines without a tokenizer
+            return max(1, len(text) // 4)
+
     def __call__(self, texts: List[str]):
         texts = list(texts)
         if self._cache is None:
@@ -143,7 +151,7 @@ class EmbedFacade:
                 out[i] = vec
                 self._cache.put(texts[i], vec)
             self.embedded_tokens += sum(
-                max(1, len(texts[i]) // 4) for i in missing)
+                self._count(texts[i]) for i in missing)
         return out
 
     @property
@@ -151,7 +159,7 @@ class EmbedFacade:
         return hasattr(self._engine, "similarity_matrix")
 
     def similarity_matrix(self, texts: List[str]):
-        self.embedded_tokens += sum(max(1, len(t) // 4) for t in texts)
+        self.embedded_tokens += sum(self._count(t) for t in texts)
         if self.has_similarity:
             return self._engine.similarity_matrix(list(texts))
         from ..consensus.rules import cosine_similarity

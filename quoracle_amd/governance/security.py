@@ -40,13 +40,19 @@ def generate_secret_value(length: int = 32, include_symbols: bool = False,
     return "".join(pysecrets.choice(alphabet) for _ in range(length))
 
 
+class VaultKeyError(Exception):
+    """No usable vault key material (parity with the reference's hard
+    requirement on CLOAK_ENCRYPTION_KEY, table_credentials.ex:1-40)."""
+
+
 class SecretVault:
     """Authenticated encryption at rest (the reference uses AES-256-GCM via
     Cloak with CLOAK_ENCRYPTION_KEY; this image has no AES library, so the
     vault uses the stdlib equivalent construction: HMAC-SHA256 in counter
     mode as the keystream PRF + an encrypt-then-MAC tag, random 16-byte
-    nonce per value).  Key material comes from QUORACLE_VAULT_KEY (or the
-    constructor); blobs are versioned, legacy v0 XOR blobs still decrypt."""
+    nonce per value).  Key material MUST come from QUORACLE_VAULT_KEY or
+    the constructor — there is no default key.  Legacy v0 XOR blobs no
+    longer decrypt implicitly; migrate them once with migrate_v0()."""
 
     _MAGIC = b"qv2:"
 
@@ -54,15 +60,21 @@ class SecretVault:
         import os as _os
         self._store = store
         if key is None:
-            raw_key = _os.environ.get("QUORACLE_VAULT_KEY",
-                                      "quoracle-amd-vault").encode()
+            env = _os.environ.get("QUORACLE_VAULT_KEY")
+            if not env:
+                raise VaultKeyError(
+                    "vault key required: set QUORACLE_VAULT_KEY (the "
+                    "reference equally refuses to run without "
+                    "CLOAK_ENCRYPTION_KEY)")
+            raw_key = env.encode()
         elif isinstance(key, str):
             raw_key = key.encode()
         else:
             raw_key = key
+        if len(raw_key) < 8:
+            raise VaultKeyError("vault key must be at least 8 bytes")
         self._enc_key = hashlib.sha256(b"enc|" + raw_key).digest()
         self._mac_key = hashlib.sha256(b"mac|" + raw_key).digest()
-        self._legacy_key = raw_key
 
     def _keystream_xor(self, nonce: bytes, data: bytes) -> bytes:
         import hmac as _hmac
@@ -86,9 +98,9 @@ class SecretVault:
     def _unseal(self, blob: bytes) -> bytes:
         import hmac as _hmac
         if not blob.startswith(self._MAGIC):
-            # legacy v0: repeating-XOR blobs from older stores
-            key = self._legacy_key
-            return bytes(b ^ key[i % len(key)] for i, b in enumerate(blob))
+            raise SecretNotFoundError(
+                "vault_format: unversioned (v0) blob — run migrate_v0() "
+                "with the legacy key to re-seal old stores")
         nonce = blob[4:20]
         tag = blob[20:52]
         ct = blob[52:]
@@ -96,6 +108,23 @@ class SecretVault:
         if not _hmac.compare_digest(tag, expect):
             raise SecretNotFoundError("vault_tag_mismatch")
         return self._keystream_xor(nonce, ct)
+
+    def migrate_v0(self, legacy_key: bytes) -> int:
+        """One-shot migration: re-seal every legacy repeating-XOR (v0) blob
+        under the current key.  Returns the number migrated."""
+        if isinstance(legacy_key, str):
+            legacy_key = legacy_key.encode()
+        migrated = 0
+        for name in self.names():
+            raw = self._store.get_secret(name)
+            if raw is None or bytes(raw).startswith(self._MAGIC):
+                continue
+            blob = bytes(raw)
+            plain = bytes(b ^ legacy_key[i % len(legacy_key)]
+                          for i, b in enumerate(blob))
+            self._store.save_secret(name, self._seal(plain), "")
+            migrated += 1
+        return migrated
 
     def put(self, name: str, value: str, description: str = "") -> None:
         self._store.save_secret(name, self._seal(value.encode()), description)
@@ -113,6 +142,17 @@ class SecretVault:
         lowered = [t.lower() for t in terms if isinstance(t, str)]
         return [n for n in self.names()
                 if any(t in n.lower() for t in lowered)]
+
+
+def has_secret_templates(params: Any) -> bool:
+    """True if any string in the (nested) params carries {{SECRET:...}}."""
+    if isinstance(params, str):
+        return bool(SECRET_TEMPLATE_RE.search(params))
+    if isinstance(params, dict):
+        return any(has_secret_templates(v) for v in params.values())
+    if isinstance(params, list):
+        return any(has_secret_templates(v) for v in params)
+    return False
 
 
 def resolve_params(params: Any, vault: SecretVault,

@@ -62,11 +62,24 @@ class TaskRuntime:
         self.registry = registry or Registry()
         self.engines = engines or EnginePool()
         self.profiles = profiles or ProfileStore(self.store)
-        self.vault = vault or SecretVault(self.store)
+        self._vault = vault          # lazy: key material only needed on use
         self.config = config or RuntimeConfig()
         self.skills = skills or SkillLoader(self.config.skills_dir)
         self.supervisor: Any = None   # set by agent.supervisor.Supervisor
         self.extras: Dict[str, Any] = {}   # injectable test hooks (http_fn, ...)
+
+    @property
+    def vault(self) -> SecretVault:
+        """Constructed on first secret use: SecretVault refuses to exist
+        without key material (QUORACLE_VAULT_KEY or an injected vault), so
+        secret-free runs never need a key."""
+        if self._vault is None:
+            self._vault = SecretVault(self.store)
+        return self._vault
+
+    @vault.setter
+    def vault(self, value: Optional[SecretVault]) -> None:
+        self._vault = value
 
     def embed_many(self, texts):
         return self.engines.embed_many_sync(texts)

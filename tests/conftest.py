@@ -8,6 +8,11 @@ import pytest
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# The vault refuses to exist without key material (production parity with
+# the reference's CLOAK_ENCRYPTION_KEY); the suite provides one like a
+# deployment would.
+os.environ.setdefault("QUORACLE_VAULT_KEY", "pytest-suite-vault-key")
+
 
 def pytest_configure(config):
     config.addinivalue_line(

@@ -14,7 +14,7 @@ from quoracle_amd.persistence.store import Store
 
 
 def _vault():
-    return sec.SecretVault(Store(":memory:"))
+    return sec.SecretVault(Store(":memory:"), key=b"test-vault-key")
 
 
 def test_secret_template_resolution_and_audit():
@@ -104,7 +104,7 @@ def test_budget_escrow_math():
 
 def test_vault_encryption_at_rest():
     store = Store(":memory:")
-    v = sec.SecretVault(store)
+    v = sec.SecretVault(store, key=b"test-vault-key")
     v.put("tok", "supersecretvalue")
     raw = store.get_secret("tok")
     assert raw is not None and b"supersecretvalue" not in raw
@@ -183,14 +183,14 @@ def test_profile_catalog_visible_to_spawners():
 
 def test_vault_authenticated_encryption_roundtrip_and_tamper():
     store = Store(":memory:")
-    v = sec.SecretVault(store, key=b"k1")
+    v = sec.SecretVault(store, key=b"key-one-0123")
     v.put("tok", "super-secret-value-123")
     raw = bytes(store.get_secret("tok"))
     assert raw.startswith(b"qv2:")
     assert b"super-secret" not in raw
     assert v.get("tok") == "super-secret-value-123"
     # distinct nonces: sealing the same value twice differs
-    v2 = sec.SecretVault(store, key=b"k1")
+    v2 = sec.SecretVault(store, key=b"key-one-0123")
     v2.put("tok2", "super-secret-value-123")
     assert bytes(store.get_secret("tok2")) != raw
     # tampering is detected
@@ -198,20 +198,34 @@ def test_vault_authenticated_encryption_roundtrip_and_tamper():
     with pytest.raises(sec.SecretNotFoundError):
         v.get("tok")
     # wrong key fails the tag check
-    v3 = sec.SecretVault(store, key=b"other")
-    store.save_secret("tok3", sec.SecretVault(store, key=b"k1")._seal(b"x"))
+    v3 = sec.SecretVault(store, key=b"other-key-456")
+    store.save_secret("tok3", sec.SecretVault(store, key=b"key-one-0123")._seal(b"x"))
     with pytest.raises(sec.SecretNotFoundError):
         v3.get("tok3")
 
 
-def test_vault_legacy_blobs_still_decrypt():
+def test_vault_refuses_to_run_without_key(monkeypatch):
+    monkeypatch.delenv("QUORACLE_VAULT_KEY", raising=False)
+    with pytest.raises(sec.VaultKeyError):
+        sec.SecretVault(Store(":memory:"))
+    with pytest.raises(sec.VaultKeyError):
+        sec.SecretVault(Store(":memory:"), key=b"short")
+
+
+def test_vault_v0_blobs_need_explicit_migration():
+    """Legacy XOR (v0) blobs no longer decrypt implicitly; migrate_v0
+    re-seals them under the real key, after which get() works."""
     store = Store(":memory:")
     legacy_key = b"quoracle-amd-vault"
     blob = bytes(b ^ legacy_key[i % len(legacy_key)]
                  for i, b in enumerate(b"oldsecret"))
     store.save_secret("old", blob)
-    v = sec.SecretVault(store, key=legacy_key)
+    v = sec.SecretVault(store, key=b"fresh-production-key")
+    with pytest.raises(sec.SecretNotFoundError):
+        v.get("old")
+    assert v.migrate_v0(legacy_key) == 1
     assert v.get("old") == "oldsecret"
+    assert bytes(store.get_secret("old")).startswith(b"qv2:")
 
 
 def test_every_action_schema_renders_and_validates_shape():
@@ -281,7 +295,7 @@ def test_vault_unseal_total_over_corrupt_blobs():
 
     from quoracle_amd.governance.security import (SecretNotFoundError,
                                                   SecretVault)
-    vault = SecretVault(_MemStore(), key=b"k1")
+    vault = SecretVault(_MemStore(), key=b"key-one-0123")
     vault.put("s", "super-secret-value")
     blob = bytearray(vault._store.d["s"])
     rng = random.Random(7)
@@ -304,7 +318,7 @@ def test_vault_unseal_total_over_corrupt_blobs():
         except (SecretNotFoundError, UnicodeDecodeError):
             pass
     # wrong key fails the tag
-    vault2 = SecretVault(_MemStore(), key=b"other")
+    vault2 = SecretVault(_MemStore(), key=b"other-key-456")
     vault2._store.d["s"] = bytes(blob)
     try:
         vault2.get("s")

@@ -33,26 +33,39 @@ async def test_messages_during_cycle_batch_into_next():
 
 @pytest.mark.asyncio
 async def test_dismiss_vs_spawn_race_guard():
-    """A spawn targeting a child id already being dismissed is refused
-    (reference: spawn.ex:76-97 dismissing flag)."""
+    """A spawn attempted while the parent has a dismissal in flight is
+    refused (reference: spawn.ex:73-107 returns :parent_dismissing).  The
+    race is created for real: dismiss_child_action marks the child as
+    dismissing synchronously and finishes in a background task, so a spawn
+    issued before that task runs hits the open window."""
     engine = FakeEngine(default_response=IDLE)
     manager, runtime = make_manager(engine)
     result = await manager.create_task("race", "default")
     root = runtime.registry.lookup(result["root_agent_id"]).actor
-    # simulate an in-progress dismissal claiming every future child id
-    import quoracle_amd.utils.ids as ids_mod
-    orig = ids_mod.agent_id
-    try:
-        ids_mod.agent_id = lambda prefix="agent": "fixed-child-id"
-        root.state.dismissing.add("fixed-child-id")
-        res = await manager.supervisor.spawn_child_action(root, {
-            "task_description": "t", "success_criteria": "s",
-            "immediate_context": "c", "approach_guidance": "a",
-            "profile": "default"})
-        assert res.get("error") == "dismissing"
-        assert "fixed-child-id" not in root.state.children
-    finally:
-        ids_mod.agent_id = orig
+    spawn_params = {"task_description": "t", "success_criteria": "s",
+                    "immediate_context": "c", "approach_guidance": "a",
+                    "profile": "default"}
+    first = await manager.supervisor.spawn_child_action(root, spawn_params)
+    child_id = first["child_id"]
+    assert child_id in root.state.children
+
+    # dismissal marked synchronously; background teardown has NOT run yet
+    res = await manager.supervisor.dismiss_child_action(root, child_id, "r")
+    assert res.get("status") == "dismissing"
+    assert root.state.dismissing
+
+    blocked = await manager.supervisor.spawn_child_action(root, spawn_params)
+    assert blocked.get("error") == "parent_dismissing"
+    assert len(root.state.children) <= 1  # no second child appeared
+
+    # once the dismissal completes, spawning works again
+    for _ in range(50):
+        if not root.state.dismissing:
+            break
+        await asyncio.sleep(0.01)
+    assert not root.state.dismissing
+    after = await manager.supervisor.spawn_child_action(root, spawn_params)
+    assert "error" not in after
     await manager.supervisor.terminate_tree(root.state.agent_id)
 
 
