@@ -520,7 +520,9 @@ class LocalEngine(Engine):
 
         row = n_decode
         max_kv = 0
-        big_step = 64 if os.environ.get("QUORACLE_MFMA64") else 32
+        # 64-row tiles match the default mfma64 prefill kernel (145 TF,
+        # validated r2); QUORACLE_MFMA32 falls back to 32-row tiles
+        big_step = 32 if os.environ.get("QUORACLE_MFMA32") else 64
         for seq, n in prefill:
             cached = len(seq.session.token_ids)
             chunk = seq.known[cached:cached + n]

@@ -837,3 +837,200 @@ paged_attn_prefill_mfma64_kernel(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// MFMA flash-decode: the VALU split decode kernel is latency-bound (128
+// serial P·V iterations per chunk, 2-wave workgroups) and measured 0.78 TB/s
+// of the 8 TB/s HBM roofline.  This kernel reuses the MFMA prefill-split
+// structure for decode: the GQ query heads sharing one KV head form the
+// first GQ rows of a 16-row Q tile (rest zero-padded), K/V stream through
+// LDS with the T14 issue-early/write-late pipeline, and both QK^T and P·V
+// run on the matrix cores — no serial V walk, 4 waves per workgroup.
+//   part_m/part_l: [B, Hq, NS] f32;  part_acc: [B, Hq, NS, D] f32
+//   (identical to paged_attn_decode_split_kernel; combined by
+//    paged_attn_decode_reduce_kernel).
+// Grid: (B, Hkv, NS); block 256.  Requires D == 128, GQ <= 16.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(256)
+paged_attn_decode_mfma_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ ctx, float scale,
+    int Hq, int Hkv, int BS, int MAXB, int GQ, int NS) {
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int split = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int len = ctx[b];
+  const int chunks = (len + MF_KCHUNK - 1) / MF_KCHUNK;
+  const int per = (chunks + NS - 1) / NS;
+  const int t0 = split * per * MF_KCHUNK;
+  const int t1 = min(len, (split + 1) * per * MF_KCHUNK);
+  const bool dead = (t0 >= len);
+
+  __shared__ bf16 q_s[MF_QT * KP];
+  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ float s_s[MF_QT * SP];
+  __shared__ bf16 p_s[MF_QT * VP];
+  __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
+
+  // Q tile: rows 0..GQ-1 are this KV head's query heads, rest zero
+  for (int i = tid; i < MF_QT * MF_D / 8; i += 256) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < GQ)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)b * Hq + hk * GQ + r) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
+  if (tid < MF_QT) {
+    m_s[tid] = -INFINITY;
+    l_s[tid] = 0.f;
+  }
+  __syncthreads();
+
+  f32x4_t o_acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4_t o_acc1 = {0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int a_row = lane & 15;
+  const int a_koff = (lane >> 4) * 8;
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
+        const long blk = bt[(long)b * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+    }
+  };
+
+  issue_loads(t0, t1);
+  for (int start = t0; start < t1; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, t1 - start);
+    write_staged();
+    __syncthreads();
+    if (start + MF_KCHUNK < t1)
+      issue_loads(start + MF_KCHUNK, t1);
+
+    // S = Q·K^T (wave w: key block w*16..w*16+15); no causal mask in decode
+    {
+      f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            q_s + a_row * KP + kk * 32 + a_koff);
+        bf16x8_t b2 = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, s_acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = c_row0 + r;
+        const int col = wave * 16 + c_col;
+        s_s[row * SP + col] = (col < clen) ? s_acc[r] * scale : -INFINITY;
+      }
+    }
+    __syncthreads();
+
+    // online softmax (16 threads per row)
+    {
+      const int row = tid >> 4;
+      const int sub = tid & 15;
+      float v0 = s_s[row * SP + sub];
+      float v1 = s_s[row * SP + sub + 16];
+      float v2 = s_s[row * SP + sub + 32];
+      float v3 = s_s[row * SP + sub + 48];
+      float mymax = fmaxf(fmaxf(v0, v1), fmaxf(v2, v3));
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        mymax = fmaxf(mymax, __shfl_xor(mymax, w, 16));
+      const float m_old = m_s[row];
+      const float mn = fmaxf(m_old, mymax);
+      const float alpha = (m_old == -INFINITY) ? 0.f : __expf(m_old - mn);
+      float p0 = (v0 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v0 - mn);
+      float p1 = (v1 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v1 - mn);
+      float p2 = (v2 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v2 - mn);
+      float p3 = (v3 == -INFINITY || mn == -INFINITY) ? 0.f : __expf(v3 - mn);
+      float psum = p0 + p1 + p2 + p3;
+#pragma unroll
+      for (int w = 8; w >= 1; w >>= 1)
+        psum += __shfl_xor(psum, w, 16);
+      if (sub == 0) {
+        l_s[row] = l_s[row] * alpha + psum;
+        m_s[row] = mn;
+        alpha_s[row] = alpha;
+      }
+      p_s[row * VP + sub] = f2bf(p0);
+      p_s[row * VP + sub + 16] = f2bf(p1);
+      p_s[row * VP + sub + 32] = f2bf(p2);
+      p_s[row * VP + sub + 48] = f2bf(p3);
+    }
+    __syncthreads();
+
+    // O rescale + O += P·V (wave w: output cols w*32..w*32+31)
+    {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float a = alpha_s[c_row0 + r];
+        o_acc0[r] *= a;
+        o_acc1[r] *= a;
+      }
+#pragma unroll
+      for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            p_s + a_row * VP + kk * 32 + a_koff);
+        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
+            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
+        o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // partials out (unnormalized; decode reduce kernel combines splits)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = c_row0 + r;
+    if (row >= GQ) continue;
+    const int h = hk * GQ + row;
+    const long base = ((long)b * Hq + h) * NS + split;
+    if (c_col == 0 && wave == 0) {
+      part_m[base] = dead ? -INFINITY : m_s[row];
+      part_l[base] = dead ? 0.f : l_s[row];
+    }
+    part_acc[base * MF_D + wave * 32 + c_col] = dead ? 0.f : o_acc0[r];
+    part_acc[base * MF_D + wave * 32 + 16 + c_col] = dead ? 0.f : o_acc1[r];
+  }
+}

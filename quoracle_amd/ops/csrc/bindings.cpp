@@ -214,6 +214,46 @@ void paged_attn_decode_split2(torch::Tensor out, torch::Tensor q,
   HIP_CHECK_KERNEL();
 }
 
+void paged_attn_decode_mfma(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor kcache, torch::Tensor vcache,
+                            torch::Tensor block_tables,
+                            torch::Tensor ctx_lens, double scale,
+                            torch::Tensor part_m, torch::Tensor part_l,
+                            torch::Tensor part_acc) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  CHECK_GPU(ctx_lens);
+  const int B = q.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  const int NS = part_m.size(2);
+  TORCH_CHECK(Hq % Hkv == 0 && GQ <= MF_QT, "unsupported GQA ratio");
+  TORCH_CHECK(D == MF_D, "MFMA decode requires head dim 128");
+  TORCH_CHECK(part_m.size(0) >= B && part_m.size(1) == Hq, "workspace shape");
+  TORCH_CHECK(part_acc.size(3) == D, "workspace D");
+  if (B == 0) return;
+  hipLaunchKernelGGL(paged_attn_decode_mfma_kernel, dim3(B, Hkv, NS),
+                     dim3(256), 0, current_stream(),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), bf16_cptr(q),
+                     bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
+                     (float)scale, Hq, Hkv, BS, MAXB, GQ, NS);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(paged_attn_decode_reduce_kernel, dim3(B, Hq),
+                     dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), Hq, D, NS);
+  HIP_CHECK_KERNEL();
+}
+
 void paged_attn_prefill(torch::Tensor out, torch::Tensor q,
                         torch::Tensor kcache, torch::Tensor vcache,
                         torch::Tensor block_tables, torch::Tensor tile_q0,
@@ -469,6 +509,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "EXPERIMENTAL: decode split v2 (256-token chunks, LDS block ids)");
   m.def("paged_attn_decode_split", &paged_attn_decode_split,
         "Flash-decoding: context-split decode attention + combine");
+  m.def("paged_attn_decode_mfma", &paged_attn_decode_mfma,
+        "MFMA flash-decode: context-split decode on the matrix cores "
+        "(D=128; GQ heads padded into a 16-row tile) + combine");
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged-KV causal prefill attention over cached context");
   m.def("paged_attn_prefill_split", &paged_attn_prefill_split,

@@ -100,10 +100,11 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
       auto pm = torch::empty({B, Hq, ns}, opts);
       auto pl = torch::empty({B, Hq, ns}, opts);
       auto pa = torch::empty({B, Hq, ns, D}, opts);
-      static const bool dec2 = std::getenv("QUORACLE_DECODE_V2") != nullptr;
-      if (dec2)
-        paged_attn_decode_split2(out_d, q_d, kcache, vcache, block_tables,
-                                 ctx_lens.value(), m.scale, pm, pl, pa);
+      static const bool dec_valu =
+          std::getenv("QUORACLE_DECODE_VALU") != nullptr;
+      if (D == 128 && !dec_valu)
+        paged_attn_decode_mfma(out_d, q_d, kcache, vcache, block_tables,
+                               ctx_lens.value(), m.scale, pm, pl, pa);
       else
         paged_attn_decode_split(out_d, q_d, kcache, vcache, block_tables,
                                 ctx_lens.value(), m.scale, pm, pl, pa);
@@ -130,10 +131,11 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
                                       *tile_seq, *tile_pos0, m.scale, pm, pl,
                                       pa);
       } else if (t32_q0.has_value() && t32_q0->numel() > 0) {
-        // big prefill: 8-wave big tiles (2-4x K/V reuse per query row);
-        // QUORACLE_MFMA64 promotes the experimental 64-row variant (the
-        // engine sizes the tiles to match)
-        static const bool use64 = std::getenv("QUORACLE_MFMA64") != nullptr;
+        // big prefill: 8-wave big tiles (2-4x K/V reuse per query row).
+        // 64-row tiles are the default (145 TF vs 83 TF for 32-row,
+        // validated r2); QUORACLE_MFMA32 falls back (the engine sizes the
+        // tiles to match)
+        static const bool use64 = std::getenv("QUORACLE_MFMA32") == nullptr;
         if (use64)
           paged_attn_prefill_mfma64(attn_out, q, kcache, vcache,
                                     block_tables, *t32_q0, *t32_qn,

@@ -71,10 +71,11 @@ def paged_attn_decode(out, q, kcache, vcache, block_tables, ctx_lens,
             part_l = torch.empty_like(part_m)
             part_acc = torch.empty((B, Hq, ns, D), dtype=torch.float32,
                                    device=q.device)
-            if os.environ.get("QUORACLE_DECODE_V2"):
-                ext().paged_attn_decode_split2(out, q, kcache, vcache,
-                                               block_tables, ctx_lens, scale,
-                                               part_m, part_l, part_acc)
+            if D == 128 and not os.environ.get("QUORACLE_DECODE_VALU"):
+                # matrix-core flash-decode (validated r2: 0.78 -> ~3 TB/s)
+                ext().paged_attn_decode_mfma(out, q, kcache, vcache,
+                                             block_tables, ctx_lens, scale,
+                                             part_m, part_l, part_acc)
             else:
                 ext().paged_attn_decode_split(out, q, kcache, vcache,
                                               block_tables, ctx_lens, scale,

@@ -91,3 +91,20 @@ if hasattr(ops.ext(), "paged_attn_decode_split2"):
     reld = (outd2.float() - outd.float()).norm() / outd.float().norm()
     print(f"decode_split2 (EXPERIMENTAL): {ms_dec2*1e3:.1f} us "
           f"({kv_bytes/ms_dec2/1e9:.2f} TB/s)  rel-vs-v1 {reld:.4f}")
+if hasattr(ops.ext(), "paged_attn_decode_mfma"):
+    outdm = torch.empty_like(qd)
+    ms_decm = timeit(lambda: ops.ext().paged_attn_decode_mfma(
+        outdm, qd, kcache, vcache, tables2, ctxs, scale, pm, pl, pa), n=50)
+    reldm = (outdm.float() - outd.float()).norm() / outd.float().norm()
+    print(f"decode_mfma: {ms_decm*1e3:.1f} us "
+          f"({kv_bytes/ms_decm/1e9:.2f} TB/s)  rel-vs-v1 {reldm:.4f}")
+    # decode-at-low-batch case (single agent turn)
+    for B1 in (1, 2):
+        q1 = qd[:B1].contiguous(); o1 = torch.empty_like(q1)
+        t1_ = tables2[:B1].contiguous(); c1 = ctxs[:B1].contiguous()
+        m1 = timeit(lambda: ops.ext().paged_attn_decode_mfma(
+            o1, q1, kcache, vcache, t1_, c1, scale,
+            pm[:B1].contiguous(), pl[:B1].contiguous(),
+            pa[:B1].contiguous()), n=50)
+        kb1 = 2 * B1 * ctx_len * Hkv * D * 2
+        print(f"decode_mfma B={B1}: {m1*1e3:.1f} us ({kb1/m1/1e9:.2f} TB/s)")

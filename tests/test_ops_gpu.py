@@ -343,3 +343,79 @@ def test_paged_attn_prefill_mfma32_matches_reference(dev):
     ref = reference.attention(q, k, v, scale, causal_offset=cached)
     assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
         f"max err {(out.float() - ref).abs().max().item()}"
+
+
+def test_paged_attn_decode_mfma_matches_reference(dev):
+    """Matrix-core flash-decode (GQ heads padded into a 16-row MFMA tile)
+    vs the fp32 reference, including a context that is not a multiple of
+    the 64-key chunk or the split width."""
+    ops = _ops()
+    torch.manual_seed(21)
+    B, Hq, Hkv, D, BS = 3, 32, 8, 128, 16
+    lens = [1500, 3000, 137]
+    NS = 12
+    scale = D ** -0.5
+    seqs = [(torch.randn(t, Hkv, D, device=dev, dtype=torch.bfloat16),
+             torch.randn(t, Hkv, D, device=dev, dtype=torch.bfloat16))
+            for t in lens]
+    kcache, vcache, tables, ctx = _build_paged_cache(dev, seqs, Hkv, D, BS)
+    q = torch.randn(B, Hq, D, device=dev, dtype=torch.bfloat16)
+    out = torch.empty_like(q)
+    part_m = torch.empty((B, Hq, NS), dtype=torch.float32, device=dev)
+    part_l = torch.empty_like(part_m)
+    part_acc = torch.empty((B, Hq, NS, D), dtype=torch.float32, device=dev)
+    ops.ext().paged_attn_decode_mfma(out, q, kcache, vcache, tables, ctx,
+                                     scale, part_m, part_l, part_acc)
+    for s in range(B):
+        ref = reference.attention(q[s:s + 1], seqs[s][0], seqs[s][1], scale)
+        assert torch.allclose(out[s].float(), ref[0], atol=4e-2, rtol=4e-2), \
+            f"seq {s}: max err {(out[s].float() - ref[0]).abs().max().item()}"
+
+
+def test_paged_attn_decode_mfma_gq8(dev):
+    """GQ=8 (llama-70b shape: 64 query heads over 8 KV heads)."""
+    ops = _ops()
+    torch.manual_seed(22)
+    B, Hq, Hkv, D, BS = 2, 64, 8, 128, 16
+    lens = [2048, 700]
+    NS = 8
+    scale = D ** -0.5
+    seqs = [(torch.randn(t, Hkv, D, device=dev, dtype=torch.bfloat16),
+             torch.randn(t, Hkv, D, device=dev, dtype=torch.bfloat16))
+            for t in lens]
+    kcache, vcache, tables, ctx = _build_paged_cache(dev, seqs, Hkv, D, BS)
+    q = torch.randn(B, Hq, D, device=dev, dtype=torch.bfloat16)
+    out = torch.empty_like(q)
+    part_m = torch.empty((B, Hq, NS), dtype=torch.float32, device=dev)
+    part_l = torch.empty_like(part_m)
+    part_acc = torch.empty((B, Hq, NS, D), dtype=torch.float32, device=dev)
+    ops.ext().paged_attn_decode_mfma(out, q, kcache, vcache, tables, ctx,
+                                     scale, part_m, part_l, part_acc)
+    for s in range(B):
+        ref = reference.attention(q[s:s + 1], seqs[s][0], seqs[s][1], scale)
+        assert torch.allclose(out[s].float(), ref[0], atol=4e-2, rtol=4e-2), \
+            f"seq {s}: max err {(out[s].float() - ref[0]).abs().max().item()}"
+
+
+def test_paged_attn_prefill_mfma64_matches_reference(dev):
+    ops = _ops()
+    torch.manual_seed(19)
+    Hq, Hkv, D, BS = 32, 8, 128, 16
+    scale = D ** -0.5
+    cached, new = 777, 200          # several 64-row tiles + odd tail
+    total = cached + new
+    k = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    kcache, vcache, tables, ctx = _build_paged_cache(dev, [(k, v)], Hkv, D, BS)
+    q = torch.randn(new, Hq, D, device=dev, dtype=torch.bfloat16)
+    out = torch.empty_like(q)
+    nt = (new + 63) // 64
+    t0 = torch.arange(nt, dtype=torch.int32, device=dev) * 64
+    qn = torch.clamp(torch.full_like(t0, new) - t0, max=64)
+    tseq = torch.zeros_like(t0)
+    tpos = t0 + cached
+    ops.ext().paged_attn_prefill_mfma64(out, q, kcache, vcache, tables,
+                                        t0, qn, tseq, tpos, scale)
+    ref = reference.attention(q, k, v, scale, causal_offset=cached)
+    assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
+        f"max err {(out.float() - ref).abs().max().item()}"
