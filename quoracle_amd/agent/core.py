@@ -484,7 +484,17 @@ class AgentActor:
             if isinstance(inject.get("profile"), str):
                 spawn_profile = inject["profile"]
                 break
-        return {"spawn_profile": spawn_profile or self.state.profile}
+        ctx: Dict[str, Any] = {
+            "spawn_profile": spawn_profile or self.state.profile}
+        # live child id so dismiss_child / adjust_budget decisions target a
+        # real child instead of a placeholder
+        if self.state.children:
+            ctx["child_id"] = next(iter(self.state.children))
+        wd = (grove.get("confinement") or {}).get("working_dir") \
+            or self.runtime.config.default_working_dir
+        ctx["file_read_path"] = f"{wd}/notes.txt"
+        ctx["file_write_path"] = f"{wd}/scratch.txt"
+        return ctx
 
     def _system_prompt(self) -> str:
         if self.state.cached_system_prompt is not None:

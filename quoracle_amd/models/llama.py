@@ -267,11 +267,17 @@ class LlamaModel:
             D.rmsnorm(h, ffn, res, next_norm, cfg.rmsnorm_eps)
         return h        # final-normed hidden states [T, hidden]
 
+    # rows at or below this take the capacity-padded batched path (no host
+    # sync, hipGraph-capturable); above it the per-expert GEMMs are
+    # compute-bound and padding waste costs more than one counts sync
+    MOE_BMM_MAX_ROWS = 512
+
     def _moe_ffn(self, h: torch.Tensor, layer: Dict[str, torch.Tensor]) -> torch.Tensor:
-        """Token-sorted MoE dispatch: flatten (token, k) pairs, sort by
-        expert, run ONE contiguous GEMM slice per expert, then a single
-        weighted scatter-add — no per-expert masking/nonzero round-trips
-        (the grouped-GEMM structure; a fused kernel can drop in later)."""
+        """MoE dispatch.  Decode-size batches: capacity-padded bins + ONE
+        bmm per FFN GEMM, zero host syncs (weight streaming dominates, so
+        the padding is free).  Big prefill chunks: token-sorted contiguous
+        GEMM slices per expert with a single counts sync amortized over
+        the chunk."""
         cfg = self.lcfg     # local expert intermediate under TP
         T = h.shape[0]
         K = cfg.top_k_experts
@@ -281,6 +287,9 @@ class LlamaModel:
         expert_flat = experts.reshape(-1)                      # [T*K]
         order = torch.argsort(expert_flat, stable=True)
         token_of = order // K                                  # source token
+        if T * K <= self.MOE_BMM_MAX_ROWS:
+            return self._moe_ffn_bmm(h, layer, weights, experts, order,
+                                     token_of)
         h_sorted = h[token_of]                                 # [T*K, d]
         counts = torch.bincount(expert_flat, minlength=cfg.n_experts)
         counts_l = counts.tolist()
@@ -299,6 +308,38 @@ class LlamaModel:
         w_sorted = weights.reshape(-1)[order].unsqueeze(1)
         out = torch.zeros_like(h)
         out.index_add_(0, token_of, down_sorted * w_sorted)
+        return out
+
+    def _moe_ffn_bmm(self, h: torch.Tensor, layer: Dict[str, torch.Tensor],
+                     weights: torch.Tensor, experts: torch.Tensor,
+                     order: torch.Tensor, token_of: torch.Tensor) -> torch.Tensor:
+        """Capacity-padded batched MoE FFN: each expert gets a fixed bin of
+        T rows (an expert receives at most one row per token), tokens are
+        scattered into their bins on the GPU, both GEMMs run as one bmm
+        over [E, T, *] — no counts.cpu(), so decode steps stay sync-free
+        and capture into hipGraphs."""
+        cfg = self.lcfg
+        T, hidden = h.shape
+        K = cfg.top_k_experts
+        E = cfg.n_experts
+        S = T * K
+        expert_flat = experts.reshape(-1)
+        e_sorted = expert_flat[order]
+        counts = torch.bincount(expert_flat, minlength=E)
+        raw_off = torch.cumsum(counts, 0) - counts             # exclusive
+        pos = torch.arange(S, device=h.device) - raw_off[e_sorted]
+        idx = e_sorted * T + pos                               # bin slots
+        bins = torch.zeros((E * T, hidden), dtype=h.dtype, device=h.device)
+        bins.index_copy_(0, idx, h[token_of])
+        gu = torch.bmm(bins.view(E, T, hidden), layer["w_gate_up"])
+        act = torch.empty((E * T, cfg.intermediate), dtype=h.dtype,
+                          device=h.device)
+        D.swiglu(act, gu.view(E * T, -1))
+        down = torch.bmm(act.view(E, T, cfg.intermediate), layer["w_down"])
+        w_sorted = weights.reshape(-1)[order].unsqueeze(1)
+        vals = down.view(E * T, hidden)[idx] * w_sorted
+        out = torch.zeros_like(h)
+        out.index_add_(0, token_of, vals)
         return out
 
     def _moe_ffn_naive(self, h: torch.Tensor, layer: Dict[str, torch.Tensor]) -> torch.Tensor:

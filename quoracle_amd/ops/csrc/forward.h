@@ -163,16 +163,60 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
   }
 }
 
+// Rows at or below this take the capacity-padded batched path (zero host
+// syncs, hipGraph-capturable).  Above it (big prefill chunks) the GEMMs are
+// compute-bound and the 4x padding waste costs more than one counts sync.
+constexpr int64_t MOE_BMM_MAX_ROWS = 512;
+
+// Capacity-padded batched MoE FFN: every expert gets a fixed bin of T rows
+// (an expert can receive at most one row per token), tokens are scattered
+// into their expert bin on the GPU, and both FFN GEMMs run as ONE bmm over
+// [E, T, *].  No counts.cpu() — decode steps stay sync-free and capture
+// into hipGraphs.  Padding is free here: at decode sizes the GEMMs are
+// bound by streaming the expert weights, not by rows.
+inline torch::Tensor moe_ffn_bmm(const Model &m, const Layer &L,
+                                 torch::Tensor h, torch::Tensor weights,
+                                 torch::Tensor experts) {
+  const int64_t T = h.size(0);
+  const int64_t K = m.top_k;
+  const int64_t E = L.router.size(1);
+  const int64_t S = T * K, C = T;
+  const int64_t hidden = h.size(1);
+  auto expert_flat = experts.reshape({-1});
+  auto order = expert_flat.argsort(/*stable=*/true);
+  auto e_sorted = expert_flat.index_select(0, order);
+  auto token_of = order.div(K, "floor");
+  auto counts = at::bincount(expert_flat, /*weights=*/{}, /*minlength=*/E);
+  auto raw_off = at::cumsum(counts, 0) - counts;   // exclusive prefix
+  auto pos = at::arange(S, order.options()) -
+             raw_off.index_select(0, e_sorted);
+  auto idx = e_sorted * C + pos;                   // slot in the bins
+  auto bins = torch::zeros({E * C, hidden}, h.options());
+  bins.index_copy_(0, idx, h.index_select(0, token_of));
+  auto gu = at::bmm(bins.view({E, C, hidden}), L.w_gate_up);
+  auto act = torch::empty({E * C, m.intermediate}, h.options());
+  swiglu(act, gu.view({E * C, gu.size(2)}));
+  auto down = at::bmm(act.view({E, C, m.intermediate}), L.w_down);
+  auto w_sorted = weights.reshape({-1}).index_select(0, order).unsqueeze(1);
+  auto vals = down.view({E * C, hidden}).index_select(0, idx) * w_sorted;
+  auto out = torch::zeros_like(h);
+  out.index_add_(0, token_of, vals);
+  return out;
+}
+
 inline torch::Tensor moe_ffn(const Model &m, const Layer &L,
                              torch::Tensor h) {
-  // token-sorted dispatch (mirrors models/llama.py _moe_ffn): one
-  // contiguous GEMM slice per expert, single weighted scatter-add
   auto logits = at::matmul(h, L.router).to(torch::kFloat32);
   auto probs = at::softmax(logits, -1);
   auto topk = probs.topk(m.top_k, -1);
   auto weights = std::get<0>(topk);
   auto experts = std::get<1>(topk);
   weights = (weights / weights.sum(-1, true)).to(h.scalar_type());
+  if (h.size(0) * m.top_k <= MOE_BMM_MAX_ROWS && h.is_cuda())
+    return moe_ffn_bmm(m, L, h, weights, experts);
+  // big prefill chunks: token-sorted dispatch, one contiguous GEMM slice
+  // per expert (compute-bound — the single counts sync amortizes over the
+  // whole chunk), single weighted scatter-add
   const int64_t E = L.router.size(1);
   auto expert_flat = experts.reshape({-1});
   auto order = expert_flat.argsort(/*stable=*/true);

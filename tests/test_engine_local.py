@@ -418,3 +418,34 @@ def test_dynamic_max_tokens_policy():
         40_000 - int(30_000 * TOKEN_SAFETY_MARGIN)
     # overflow-sized input: floored at the minimum, never negative
     assert dynamic_max_tokens(eng, "m", 200_000) == MIN_OUTPUT_TOKENS
+
+
+def test_moe_bmm_path_matches_naive_and_sorted():
+    """Capacity-padded bmm MoE (decode path, no host sync) and the
+    token-sorted loop (prefill path) both reproduce the naive per-expert
+    reference."""
+    from dataclasses import replace
+    import torch
+    from quoracle_amd.models.config import PRESETS
+    from quoracle_amd.models.llama import LlamaModel
+    cfg = replace(PRESETS["tiny"], n_experts=4, top_k_experts=2)
+    torch.manual_seed(7)
+    m = LlamaModel("moe-paths", torch.device("cpu"), dtype=torch.float32,
+                   cfg=cfg)
+    layer = m.layers[0]
+    for T in (1, 5, 16):
+        h = torch.randn(T, cfg.hidden, dtype=torch.float32)
+        naive = m._moe_ffn_naive(h, layer)
+        assert T * cfg.top_k_experts <= LlamaModel.MOE_BMM_MAX_ROWS
+        bmm = m._moe_ffn(h, layer)
+        assert torch.allclose(naive, bmm, atol=1e-4), \
+            f"T={T}: {(naive - bmm).abs().max().item()}"
+    old = LlamaModel.MOE_BMM_MAX_ROWS
+    try:
+        LlamaModel.MOE_BMM_MAX_ROWS = 0     # force the sorted-loop path
+        h = torch.randn(9, cfg.hidden, dtype=torch.float32)
+        naive = m._moe_ffn_naive(h, layer)
+        loop = m._moe_ffn(h, layer)
+        assert torch.allclose(naive, loop, atol=1e-4)
+    finally:
+        LlamaModel.MOE_BMM_MAX_ROWS = old

@@ -191,3 +191,32 @@ def test_tokenizer_edges():
     assert tok.decode([]) == ""
     assert tok.decode([EOS]) == ""          # EOS renders empty
     assert tok.count("héllo") == len("héllo".encode())
+
+
+def test_grammar_covers_all_22_actions_and_validates():
+    """The constrained-decoding grammar templates every action in the
+    schema registry; a forced walk through each choice yields JSON that
+    parses and passes schema validation."""
+    import json
+    import random
+    from quoracle_amd.engine.sampler import ActionGrammar, _PARAM_PLANS
+    from quoracle_amd.engine.tokenizer import EOS
+    from quoracle_amd.actions.schema import ACTIONS
+    from quoracle_amd.actions.validator import validate_params
+
+    assert set(_PARAM_PLANS) == set(ACTIONS)
+    rng = random.Random(11)
+    for action in ACTIONS:
+        idx = ACTIONS.index(action)
+        for _ in range(3):
+            g = ActionGrammar(ACTIONS, context={"child_id": "kid-1"})
+            out = []
+            while not g.done and len(out) < 4000:
+                op = g.current()
+                tok = g.advance(idx if op[0] == "choice"
+                                else rng.randrange(0, 256))
+                out.append(tok)
+            assert g.done, f"{action}: grammar never terminated"
+            doc = json.loads(bytes(b for b in out if b != EOS).decode())
+            assert doc["action"] == action
+            validate_params(action, doc.get("params", {}))
