@@ -9,9 +9,11 @@ only synchronization is an admit-broadcast at the top of every engine step
 result futures; replicas execute identical forwards so the per-layer RCCL
 all-reduces line up (BASELINE config 5).
 
-Known limitation: the single-engine fault watchdog (engine.reset_model)
-is rank-local; a device fault on one TP rank desynchronizes the group and
-requires a group-level restart (tracked in ROADMAP.md).
+Group-level watchdog: a device fault on one rank defers its KV reset to
+the next step's synchronization point, where a control-plane MAX
+all-reduce of the pending-reset mask makes EVERY rank rebuild the same
+models together — the lockstep invariant (identical block allocation,
+identical batch composition) survives the reset.
 
 Exactness requirements (hold by construction):
   * identical model shards from the same seed (models/llama.py TP sharding),
@@ -56,13 +58,43 @@ class TPEngine(LocalEngine):
         # step (a leader admitting unannounced work would desync the
         # per-layer all-reduce counts and deadlock the group)
         self._tp_direct = True
+        # group watchdog: model keys pending a synchronized reset
+        self._reset_pending: set = set()
+        self._model_order = sorted(self.models)
 
     def request_stop(self) -> None:
         """Leader: broadcast shutdown to replicas on the next step."""
         self._inbox.put(_StopSentinel())
 
+    def reset_model(self, key: str) -> None:
+        """Group-level watchdog: a rank-local fault must not reset one
+        replica alone (block allocation would diverge and the per-layer
+        all-reduces deadlock).  Defer to the next step's sync point where
+        the whole group resets together."""
+        if self.tp.world > 1:
+            self._reset_pending.add(key)
+            return
+        super().reset_model(key)
+
+    def _sync_group_resets(self) -> None:
+        """MAX-all-reduce the pending-reset mask over the control group and
+        apply the union on every rank at the same step boundary."""
+        mask = torch.zeros(len(self._model_order), dtype=torch.int32)
+        for i, key in enumerate(self._model_order):
+            if key in self._reset_pending:
+                mask[i] = 1
+        dist.all_reduce(mask, op=dist.ReduceOp.MAX, group=self.control_group)
+        self._reset_pending.clear()
+        for i, key in enumerate(self._model_order):
+            if int(mask[i]):
+                super().reset_model(key)
+
     def step(self) -> bool:
         if self.tp.world > 1:
+            # group watchdog first: apply any rank's pending resets
+            # everywhere BEFORE admitting this step's requests, so fresh
+            # work is never killed by a reset it arrived after
+            self._sync_group_resets()
             if self.is_leader:
                 drained = self._drain_inbox()
                 stop = any(isinstance(s, _StopSentinel) for s in drained)

@@ -66,3 +66,60 @@ def test_tp_engine_lockstep_matches_single():
     assert err is None, err
     assert all(p.exitcode == 0 for p in procs)
     assert text == single, f"TP text diverged:\n{text}\nvs\n{single}"
+
+
+def _tp_reset_worker(rank, world, port, out_q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from quoracle_amd.engine.api import GenerateRequest
+        from quoracle_amd.engine.tp_engine import TPEngine, serve_tp_replica
+        from quoracle_amd.parallel.tp import TPContext
+        tp = TPContext(rank, world)
+        eng = TPEngine(["tiny#tp"], tp, device=torch.device("cpu"),
+                       kv_blocks_override=256, embed_model_key=None,
+                       prefill_chunk=64)
+        mgr_before = id(eng.models["tiny#tp"].mgr)
+        if rank == 1:
+            # rank-local fault: must NOT reset immediately (would desync)
+            eng.reset_model("tiny#tp")
+            assert id(eng.models["tiny#tp"].mgr) == mgr_before
+            assert eng._reset_pending == {"tiny#tp"}
+        if rank == 0:
+            r = eng.generate_sync(GenerateRequest(
+                model_key="tiny#tp",
+                messages=[{"role": "user", "content": "after fault"}],
+                temperature=0.7, max_tokens=200, seed=5,
+                action_grammar=True, session_id="tpr"), timeout=300)
+            eng.request_stop()
+            eng.step()
+            out_q.put(("r0", r.error, id(eng.models["tiny#tp"].mgr) != mgr_before))
+        else:
+            serve_tp_replica(eng)
+            out_q.put(("r1", None, id(eng.models["tiny#tp"].mgr) != mgr_before))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_tp_group_watchdog_resets_all_ranks_together():
+    """A fault on ONE TP rank defers its KV reset to the step sync point,
+    where the MAX-all-reduced mask makes EVERY rank rebuild — lockstep
+    block allocation survives and generation still completes."""
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_reset_worker, args=(r, 2, 29547, out_q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        who, err, reset_applied = out_q.get(timeout=240)
+        results[who] = (err, reset_applied)
+    for p in procs:
+        p.join(timeout=120)
+    assert all(p.exitcode == 0 for p in procs)
+    assert results["r0"][0] is None, results["r0"][0]
+    # BOTH ranks rebuilt their BlockManager (not just the faulty one)
+    assert results["r0"][1] and results["r1"][1], results
