@@ -427,6 +427,87 @@ class MCPConnection:
         msg = {"jsonrpc": "2.0", "method": method, "params": params or {}}
         self.proc.stdin.write((json.dumps(msg) + "\n").encode())
 
+    async def close(self) -> None:
+        self.proc.terminate()
+        try:
+            await asyncio.wait_for(self.proc.wait(), 5)
+        except (asyncio.TimeoutError, ProcessLookupError):
+            self.proc.kill()
+
+
+class MCPHttpConnection:
+    """Streamable-HTTP MCP client (reference: actions/mcp.ex:48-104
+    supports stdio AND streamable HTTP).  JSON-RPC over POST; the server
+    may answer application/json or a text/event-stream carrying the JSON
+    message; an Mcp-Session-Id response header is echoed on later calls
+    and released with DELETE on close."""
+
+    def __init__(self, connection_id: str, url: str):
+        import httpx
+        self.connection_id = connection_id
+        self.url = url
+        self.proc = None
+        self._next_id = 0
+        self.tools: List[dict] = []
+        self._session_id: Optional[str] = None
+        self._client = httpx.AsyncClient()
+
+    def _headers(self) -> dict:
+        h = {"Content-Type": "application/json",
+             "Accept": "application/json, text/event-stream"}
+        if self._session_id:
+            h["Mcp-Session-Id"] = self._session_id
+        return h
+
+    @staticmethod
+    def _from_sse(text: str, want_id: int) -> Any:
+        for line in text.splitlines():
+            if not line.startswith("data:"):
+                continue
+            try:
+                msg = json.loads(line[5:].strip())
+            except json.JSONDecodeError:
+                continue
+            if msg.get("id") == want_id:
+                return msg
+        raise ConnectionError("no matching message in SSE stream")
+
+    async def rpc(self, method: str, params: Optional[dict] = None,
+                  timeout_s: float = 30.0) -> Any:
+        self._next_id += 1
+        msg = {"jsonrpc": "2.0", "id": self._next_id, "method": method,
+               "params": params or {}}
+        resp = await self._client.post(self.url, json=msg,
+                                       headers=self._headers(),
+                                       timeout=timeout_s)
+        if resp.status_code >= 400:
+            raise ConnectionError(f"http {resp.status_code}: {resp.text[:200]}")
+        sid = resp.headers.get("mcp-session-id")
+        if sid:
+            self._session_id = sid
+        ctype = resp.headers.get("content-type", "")
+        if "text/event-stream" in ctype:
+            reply = self._from_sse(resp.text, self._next_id)
+        else:
+            reply = resp.json()
+        if "error" in reply:
+            raise RuntimeError(json.dumps(reply["error"]))
+        return reply.get("result")
+
+    def notify(self, method: str, params: Optional[dict] = None) -> None:
+        msg = {"jsonrpc": "2.0", "method": method, "params": params or {}}
+        asyncio.ensure_future(self._client.post(
+            self.url, json=msg, headers=self._headers(), timeout=10.0))
+
+    async def close(self) -> None:
+        try:
+            if self._session_id:
+                await self._client.delete(self.url, headers=self._headers(),
+                                          timeout=5.0)
+        except Exception:  # noqa: BLE001 — session release is best-effort
+            pass
+        await self._client.aclose()
+
 
 async def execute_call_mcp(ctx) -> Dict[str, Any]:
     params = ctx.params
@@ -438,11 +519,7 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
         if conn is None:
             return _err("unknown_connection_id")
         if params.get("terminate"):
-            conn.proc.terminate()
-            try:
-                await asyncio.wait_for(conn.proc.wait(), 5)
-            except (asyncio.TimeoutError, ProcessLookupError):
-                conn.proc.kill()
+            await conn.close()
             agent.mcp_connections.pop(params["connection_id"], None)
             return {"connection_id": conn.connection_id, "status": "terminated"}
         tool = params.get("tool")
@@ -479,18 +556,31 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
             tools_result = await conn.rpc("tools/list", {}, timeout_s)
             conn.tools = (tools_result or {}).get("tools", [])
         except Exception as exc:  # noqa: BLE001
-            proc.terminate()
-            try:
-                await asyncio.wait_for(proc.wait(), 5)
-            except (asyncio.TimeoutError, ProcessLookupError):
-                proc.kill()
+            await conn.close()
             return _err("mcp_connect_failed", detail=str(exc))
         agent.mcp_connections[conn.connection_id] = conn
         return {"connection_id": conn.connection_id,
                 "tools": conn.tools, "status": "connected"}
     if transport == "http":
-        return _err("http_transport_unavailable",
-                    detail="HTTP MCP transport requires network egress")
+        url = params.get("url")
+        if not url:
+            return _err("missing_url")
+        conn = MCPHttpConnection(ids.connection_id(), url)
+        try:
+            await conn.rpc("initialize", {
+                "protocolVersion": "2024-11-05",
+                "capabilities": {},
+                "clientInfo": {"name": "quoracle-amd", "version": "0.1"}},
+                timeout_s)
+            conn.notify("notifications/initialized")
+            tools_result = await conn.rpc("tools/list", {}, timeout_s)
+            conn.tools = (tools_result or {}).get("tools", [])
+        except Exception as exc:  # noqa: BLE001
+            await conn.close()
+            return _err("mcp_connect_failed", detail=str(exc))
+        agent.mcp_connections[conn.connection_id] = conn
+        return {"connection_id": conn.connection_id,
+                "tools": conn.tools, "status": "connected"}
     return _err("invalid_transport")
 
 

@@ -116,3 +116,86 @@ async def test_answer_engine_uses_configured_role_model():
                                       {"prompt": "capital of France?"}))
     assert "Paris" in json.dumps(res)
     assert res.get("model") == "fake-b" or "fake-b" in json.dumps(res)
+
+
+@pytest.mark.asyncio
+async def test_call_mcp_http_roundtrip():
+    """Full MCP lifecycle over streamable HTTP against a REAL local server
+    (reference: actions/mcp.ex:48-104 stdio + streamable HTTP parity).
+    The server answers initialize as SSE (testing the event-stream parse
+    path), later calls as plain JSON, assigns a session id, and records
+    the DELETE on terminate."""
+    import socket
+    import threading
+    import http.server
+
+    state = {"deleted": False, "session_seen": []}
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):   # noqa: D102 — silence test output
+            pass
+
+        def do_POST(self):
+            body = json.loads(self.rfile.read(
+                int(self.headers["Content-Length"])))
+            state["session_seen"].append(self.headers.get("Mcp-Session-Id"))
+            if "id" not in body:      # notification
+                self.send_response(202)
+                self.end_headers()
+                return
+            if body["method"] == "initialize":
+                reply = {"jsonrpc": "2.0", "id": body["id"],
+                         "result": {"serverInfo": {"name": "t"}}}
+                payload = f"data: {json.dumps(reply)}\n\n".encode()
+                self.send_response(200)
+                self.send_header("Content-Type", "text/event-stream")
+                self.send_header("Mcp-Session-Id", "sess-42")
+                self.send_header("Content-Length", str(len(payload)))
+                self.end_headers()
+                self.wfile.write(payload)
+                return
+            if body["method"] == "tools/list":
+                result = {"tools": [{"name": "add"}]}
+            elif body["method"] == "tools/call":
+                args = body["params"]["arguments"]
+                result = {"content": [{"type": "text",
+                                       "text": str(args["a"] + args["b"])}]}
+            else:
+                result = {}
+            payload = json.dumps({"jsonrpc": "2.0", "id": body["id"],
+                                  "result": result}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(payload)))
+            self.end_headers()
+            self.wfile.write(payload)
+
+        def do_DELETE(self):
+            state["deleted"] = True
+            self.send_response(200)
+            self.end_headers()
+
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+    port = srv.server_address[1]
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        runtime = make_runtime()
+        actor = _actor(runtime)
+        res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+            "transport": "http", "url": f"http://127.0.0.1:{port}/mcp"}))
+        assert res.get("status") == "connected", res
+        assert res["tools"] == [{"name": "add"}]
+        cid = res["connection_id"]
+        res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+            "connection_id": cid, "tool": "add",
+            "arguments": {"a": 2, "b": 3}}))
+        assert "5" in json.dumps(res), res
+        res = await R.execute_action(_ctx(actor, runtime, "call_mcp", {
+            "connection_id": cid, "terminate": True}))
+        assert res.get("status") == "terminated"
+        assert state["deleted"], "session DELETE not sent on terminate"
+        # session id echoed on post-initialize calls
+        assert "sess-42" in state["session_seen"]
+    finally:
+        srv.shutdown()
