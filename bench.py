@@ -35,6 +35,8 @@ def parse_args():
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--pool-size", type=int, default=3)
     p.add_argument("--agents-per-gpu", type=int, default=8)
+    p.add_argument("--rounds", type=int, default=4,
+                   help="max refinement rounds (reference default 4)")
     p.add_argument("--kv-gb", type=float, default=24.0)
     p.add_argument("--device", default=None, help="override (e.g. cpu)")
     p.add_argument("--pool-scope", choices=["shard", "global"],
@@ -147,7 +149,7 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
     from quoracle_amd.agent.state import AgentState
     from quoracle_amd.agent.supervisor import Supervisor
     from quoracle_amd.engine.pool import EnginePool
-    from quoracle_amd.governance.profiles import Profile
+    from quoracle_amd.governance.profiles import Profile, VALID_GROUPS
     from quoracle_amd.tasks.runtime import RuntimeConfig, TaskRuntime
     from quoracle_amd.utils import ids
 
@@ -186,7 +188,8 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
             model_pool=pool_keys or (
                 global_pool if args.pool_scope == "global"
                 else keys_for_rank(r)),
-            capability_groups=[], max_refinement_rounds=2))
+            capability_groups=list(VALID_GROUPS),
+            max_refinement_rounds=args.rounds))
     if pool_keys:
         for key in pool_keys:
             pool.assign(key, engine)
@@ -202,7 +205,7 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
             task_id=f"bench-task-{shard}", parent_id=parent_id,
             profile=f"bench-r{shard}",
             model_pool=list(profile.model_pool),
-            capability_groups=[],
+            capability_groups=list(profile.capability_groups),
             max_refinement_rounds=profile.max_refinement_rounds,
         )
         state.init_model_maps()
@@ -320,7 +323,7 @@ async def orchestrate(args, engine, device, world, pool_keys=None,
                 statistics.median(step_latencies), 2) if step_latencies else None,
             "decisions_completed": decisions,
             "constrained_decoding": True,
-            "max_refinement_rounds": 2,
+            "max_refinement_rounds": args.rounds,
             "engine_stats": dict(engine.stats),
         },
     }

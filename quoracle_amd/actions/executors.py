@@ -608,11 +608,29 @@ async def execute_answer_engine(ctx) -> Dict[str, Any]:
 
 
 async def execute_generate_images(ctx) -> Dict[str, Any]:
+    """Image generation: an injected image_fn (external model) wins;
+    otherwise the locally-hosted procedural model renders a real PNG
+    (utils/imagegen.py — the no-network stand-in for a diffusion model,
+    reference: models/image_query.ex).  Data-URL payloads flow through
+    the image-artifact pipeline into multimodal history entries."""
+    import base64 as _b64
+    prompt = ctx.params["prompt"]
+    source = ctx.params.get("source_image")
     image_fn = ctx.runtime.extras.get("image_fn")
-    if image_fn is None:
-        return _err("no_image_model_configured")
-    images = await image_fn(ctx.params["prompt"], ctx.params.get("source_image"))
-    return {"images": images}
+    if image_fn is not None:
+        images = await image_fn(prompt, source)
+        return {"images": images}
+    from ..utils import imagegen
+    src_bytes = None
+    if isinstance(source, str) and source:
+        try:
+            src_bytes = _b64.b64decode(source, validate=False)
+        except Exception:  # noqa: BLE001
+            src_bytes = source.encode()
+    png = imagegen.render(prompt, source_image=src_bytes)
+    data_url = "data:image/png;base64," + _b64.b64encode(png).decode()
+    return {"images": [data_url], "model": imagegen.MODEL_NAME,
+            "mode": "edit" if src_bytes else "generate"}
 
 
 # ---------------------------------------------------------------------------

@@ -169,6 +169,10 @@ async def execute_action(ctx: ActionContext) -> Dict[str, Any]:
         runtime.bus.log(state.agent_id, "info",
                         f"{len(image_artifacts)} image artifact(s) detected",
                         {"artifacts": image_artifacts})
+        if isinstance(result, dict):
+            # artifact metadata rides with the result (the reference's
+            # multimodal history entry carries the image reference)
+            result = {**result, "image_artifacts": image_artifacts}
 
     # 8. Broadcast + persist
     runtime.bus.action_event(state.agent_id, "completed", ctx.action, ctx.action_id,
