@@ -652,18 +652,26 @@ paged_attn_prefill_mfma64_kernel(
   const int pos0 = tile_pos0[tile];
   const int kv_limit = pos0 + qn;
 
-  // LDS: no q_s — each wave's Q A-fragments are loop-invariant and live in
-  // registers (32 VGPRs), freeing 17 KB of LDS and 8 ds_reads per chunk
+  __shared__ bf16 q_s[MF4_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
   __shared__ bf16 vt_s[MF_D * VP];
   __shared__ float s_s[MF4_QT * SP];
   __shared__ bf16 p_s[MF4_QT * VP];
   __shared__ float m_s[MF4_QT], l_s[MF4_QT], alpha_s[MF4_QT];
 
+  for (int i = tid; i < MF4_QT * MF_D / 8; i += 512) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+  }
   if (tid < MF4_QT) {
     m_s[tid] = -INFINITY;
     l_s[tid] = 0.f;
   }
+  __syncthreads();
 
   // two qblocks per wave: frags xN for qblock qb0 and qb0+1
   f32x4_t s_acc0, s_acc1;
@@ -679,23 +687,6 @@ paged_attn_prefill_mfma64_kernel(
   const int c_row0 = (lane >> 4) * 4;
   const int kblock = wave & 3;
   const int qb0 = (wave >> 2) * 2;
-
-  // Q A-fragments straight from HBM into registers (rows >= qn zeroed):
-  // lane reads q[row = qb*16 + a_row][kk*32 + a_koff .. +8]
-  bf16x8_t q_frag[2][MF_D / 32];
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-    const int row = (qb0 + half) * 16 + a_row;
-#pragma unroll
-    for (int kk = 0; kk < MF_D / 32; ++kk) {
-      bf16x8_t v = {};
-      if (row < qn)
-        v = *reinterpret_cast<const bf16x8_t *>(
-            q + ((long)(q0 + row) * Hq + h) * MF_D + kk * 32 + a_koff);
-      q_frag[half][kk] = v;
-    }
-  }
-  __syncthreads();
 
   uint4 kreg[2], vreg[2];
   auto issue_loads = [&](int start_, int limit_) {
@@ -743,10 +734,12 @@ paged_attn_prefill_mfma64_kernel(
       for (int kk = 0; kk < MF_D / 32; ++kk) {
         bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
             k_s + (kblock * 16 + a_row) * KP + kk * 32 + a_koff);
-        s_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[0][kk], b,
-                                                         s_acc0, 0, 0, 0);
-        s_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[1][kk], b,
-                                                         s_acc1, 0, 0, 0);
+        bf16x8_t a0 = *reinterpret_cast<const bf16x8_t *>(
+            q_s + ((qb0 + 0) * 16 + a_row) * KP + kk * 32 + a_koff);
+        bf16x8_t a1 = *reinterpret_cast<const bf16x8_t *>(
+            q_s + ((qb0 + 1) * 16 + a_row) * KP + kk * 32 + a_koff);
+        s_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, s_acc0, 0, 0, 0);
+        s_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, s_acc1, 0, 0, 0);
       }
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
