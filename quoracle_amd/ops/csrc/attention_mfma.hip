@@ -34,6 +34,8 @@
 #define KP (MF_D + 8)        // k_s / q_s row stride (elements)
 #define VP (MF_KCHUNK + 8)   // vt_s / p_s row stride
 #define SP (MF_KCHUNK + 4)   // s_s row stride (f32 words)
+#define VR (MF_D + 6)        // row-major V stride (67 dwords, odd: spreads
+                             // the PV key-gather across all LDS banks)
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
@@ -60,7 +62,7 @@ paged_attn_prefill_mfma_kernel(
 
   __shared__ bf16 q_s[MF_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
-  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
   __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
@@ -118,9 +120,10 @@ paged_attn_prefill_mfma_kernel(
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
       reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
     }
   };
 
@@ -203,10 +206,13 @@ paged_attn_prefill_mfma_kernel(
       for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
             p_s + a_row * VP + kk * 32 + a_koff);
-        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
-        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b0, b1;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) {
+          const bf16 *vrow = v_s + (kk * 32 + a_koff + t) * VR + wave * 32;
+          b0[t] = vrow[c_col];
+          b1[t] = vrow[16 + c_col];
+        }
         o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
         o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
       }
@@ -265,7 +271,7 @@ paged_attn_prefill_mfma_split_kernel(
 
   __shared__ bf16 q_s[MF_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
-  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
   __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
@@ -321,9 +327,10 @@ paged_attn_prefill_mfma_split_kernel(
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
       reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
     }
   };
 
@@ -402,10 +409,13 @@ paged_attn_prefill_mfma_split_kernel(
       for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
             p_s + a_row * VP + kk * 32 + a_koff);
-        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
-        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b0, b1;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) {
+          const bf16 *vrow = v_s + (kk * 32 + a_koff + t) * VR + wave * 32;
+          b0[t] = vrow[c_col];
+          b1[t] = vrow[16 + c_col];
+        }
         o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
         o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
       }
@@ -461,7 +471,7 @@ paged_attn_prefill_mfma32_kernel(
 
   __shared__ bf16 q_s[MF2_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
-  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF2_QT * SP];
   __shared__ bf16 p_s[MF2_QT * VP];
   __shared__ float m_s[MF2_QT], l_s[MF2_QT], alpha_s[MF2_QT];
@@ -517,9 +527,10 @@ paged_attn_prefill_mfma32_kernel(
       const int i = tid + it * 512;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
       reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
     }
   };
 
@@ -600,10 +611,13 @@ paged_attn_prefill_mfma32_kernel(
       for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
             p_s + (qblock * 16 + a_row) * VP + kk * 32 + a_koff);
-        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (kblock * 32 + c_col) * VP + kk * 32 + a_koff);
-        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (kblock * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b0, b1;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) {
+          const bf16 *vrow = v_s + (kk * 32 + a_koff + t) * VR + kblock * 32;
+          b0[t] = vrow[c_col];
+          b1[t] = vrow[16 + c_col];
+        }
         o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
         o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
       }
@@ -652,9 +666,15 @@ paged_attn_prefill_mfma64_kernel(
   const int pos0 = tile_pos0[tile];
   const int kv_limit = pos0 + qn;
 
+  // V stays ROW-major in LDS (key-major, stride VR elems).  The old
+  // transposed image vt_s[d][key] made the staging scatter a 32-way bank
+  // conflict (8-row lane stride x 136-elem rows = bank step 32 mod 64);
+  // row-major staging writes are conflict-free b32s, and VR = 134
+  // (67 dwords, odd) spreads the PV B-fragment's scalar key-gather over
+  // all banks: key groups land at +24k mod 64, d-columns at +c/2.
   __shared__ bf16 q_s[MF4_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
-  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ bf16 v_s[MF_KCHUNK * VR];
   __shared__ float s_s[MF4_QT * SP];
   __shared__ bf16 p_s[MF4_QT * VP];
   __shared__ float m_s[MF4_QT], l_s[MF4_QT], alpha_s[MF4_QT];
@@ -713,9 +733,13 @@ paged_attn_prefill_mfma64_kernel(
       const int i = tid + it * 512;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
       reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+      // V row-major: 4 b32 stores (VR row stride is 4-byte aligned only);
+      // conflict-free — lanes sharing a key span banks 4 apart, the four
+      // keys per instruction land at distinct bank residues (VR/2 odd)
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
     }
   };
 
@@ -804,10 +828,15 @@ paged_attn_prefill_mfma64_kernel(
       }
 #pragma unroll
       for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
-        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (kblock * 32 + c_col) * VP + kk * 32 + a_koff);
-        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (kblock * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        // B fragments gathered from row-major V: lane t-loop walks 8 keys
+        // at a fixed d column (conflict-free by VR construction)
+        bf16x8_t b0, b1;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) {
+          const bf16 *vrow = v_s + (kk * 32 + a_koff + t) * VR + kblock * 32;
+          b0[t] = vrow[c_col];
+          b1[t] = vrow[16 + c_col];
+        }
         bf16x8_t a0 = *reinterpret_cast<const bf16x8_t *>(
             p_s + ((qb0 + 0) * 16 + a_row) * VP + kk * 32 + a_koff);
         bf16x8_t a1 = *reinterpret_cast<const bf16x8_t *>(
@@ -873,7 +902,7 @@ paged_attn_decode_mfma_kernel(
 
   __shared__ bf16 q_s[MF_QT * KP];
   __shared__ bf16 k_s[MF_KCHUNK * KP];
-  __shared__ bf16 vt_s[MF_D * VP];
+  __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
   __shared__ float m_s[MF_QT], l_s[MF_QT], alpha_s[MF_QT];
@@ -927,9 +956,10 @@ paged_attn_decode_mfma_kernel(
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
       reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
-      const bf16 *ve = reinterpret_cast<const bf16 *>(&vreg[it]);
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) vt_s[(d + j) * VP + key] = ve[j];
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
     }
   };
 
@@ -1008,10 +1038,13 @@ paged_attn_decode_mfma_kernel(
       for (int kk = 0; kk < MF_KCHUNK / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
             p_s + a_row * VP + kk * 32 + a_koff);
-        bf16x8_t b0 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + c_col) * VP + kk * 32 + a_koff);
-        bf16x8_t b1 = *reinterpret_cast<const bf16x8_t *>(
-            vt_s + (wave * 32 + 16 + c_col) * VP + kk * 32 + a_koff);
+        bf16x8_t b0, b1;
+#pragma unroll
+        for (int t = 0; t < 8; ++t) {
+          const bf16 *vrow = v_s + (kk * 32 + a_koff + t) * VR + wave * 32;
+          b0[t] = vrow[c_col];
+          b1[t] = vrow[16 + c_col];
+        }
         o_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, o_acc0, 0, 0, 0);
         o_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, o_acc1, 0, 0, 0);
       }
