@@ -66,6 +66,24 @@ void swiglu(torch::Tensor out, torch::Tensor gate_up) {
   HIP_CHECK_KERNEL();
 }
 
+void split_qkv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+               torch::Tensor qkv) {
+  CHECK_GPU(q);
+  CHECK_GPU(k);
+  CHECK_GPU(v);
+  CHECK_GPU(qkv);
+  const long T = qkv.size(0);
+  const int q_dim = q.numel() / T;
+  const int kv_dim = k.numel() / T;
+  TORCH_CHECK(qkv.size(1) == q_dim + 2 * kv_dim, "qkv row layout");
+  TORCH_CHECK(q_dim % 8 == 0 && kv_dim % 8 == 0, "dims % 8");
+  if (T == 0) return;
+  hipLaunchKernelGGL(split_qkv_kernel, dim3((unsigned)T), dim3(256), 0,
+                     current_stream(), bf16_ptr(q), bf16_ptr(k), bf16_ptr(v),
+                     bf16_cptr(qkv), q_dim, kv_dim);
+  HIP_CHECK_KERNEL();
+}
+
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor pos,
                   double theta) {
   CHECK_GPU(q);
@@ -501,6 +519,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fused", &rmsnorm_fused,
         "Fused residual-add + RMSNorm (bf16, fp32 accum)");
   m.def("swiglu", &swiglu, "Fused SiLU(gate) * up");
+  m.def("split_qkv", &split_qkv,
+        "One-pass split of the fused QKV GEMM output into q/k/v");
   m.def("rope_inplace", &rope_inplace, "Rotate-half RoPE in place on q/k");
   m.def("kv_append", &kv_append, "Scatter K/V rows into the paged cache");
   m.def("paged_attn_decode", &paged_attn_decode,

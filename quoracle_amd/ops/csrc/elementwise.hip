@@ -220,3 +220,28 @@ extern "C" __global__ void gather_rows_kernel(
   uint4 *o = reinterpret_cast<uint4 *>(out + t * (long)n);
   for (int i = threadIdx.x; i < n / 8; i += blockDim.x) o[i] = s[i];
 }
+
+// ---------------------------------------------------------------------------
+// split_qkv: one pass splitting the fused QKV GEMM output into contiguous
+// q/k/v tensors (replaces three narrow().contiguous() dispatch+copy chains
+// per layer in the forward driver).  Row layout: [q_dim | kv_dim | kv_dim].
+//   qkv: [T, q_dim + 2*kv_dim] bf16;  q: [T, q_dim];  k/v: [T, kv_dim]
+// Grid: (T); block 256.  Dims % 8 == 0.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void split_qkv_kernel(
+    bf16 *__restrict__ q, bf16 *__restrict__ k, bf16 *__restrict__ v,
+    const bf16 *__restrict__ qkv, int q_dim, int kv_dim) {
+  const long t = blockIdx.x;
+  const long row = t * (long)(q_dim + 2 * kv_dim);
+  const uint4 *src = reinterpret_cast<const uint4 *>(qkv + row);
+  uint4 *qo = reinterpret_cast<uint4 *>(q + t * (long)q_dim);
+  uint4 *ko = reinterpret_cast<uint4 *>(k + t * (long)kv_dim);
+  uint4 *vo = reinterpret_cast<uint4 *>(v + t * (long)kv_dim);
+  const int qv = q_dim / 8, kv8 = kv_dim / 8;
+  for (int i = threadIdx.x; i < qv + 2 * kv8; i += blockDim.x) {
+    const uint4 val = src[i];
+    if (i < qv) qo[i] = val;
+    else if (i < qv + kv8) ko[i - qv] = val;
+    else vo[i - qv - kv8] = val;
+  }
+}

@@ -271,15 +271,14 @@ inline torch::Tensor forward(
 
   auto attn_out = torch::empty({T, m.n_heads, m.head_dim}, m.embed.options());
 
+  auto q = torch::empty({T, m.n_heads, m.head_dim}, m.embed.options());
+  auto k = torch::empty({T, m.n_kv_heads, m.head_dim}, m.embed.options());
+  auto v = torch::empty({T, m.n_kv_heads, m.head_dim}, m.embed.options());
   for (size_t li = 0; li < m.layers.size(); ++li) {
     const Layer &L = m.layers[li];
     auto qkv = at::matmul(h, L.wqkv);
-    auto q = qkv.narrow(1, 0, q_dim).view({T, m.n_heads, m.head_dim})
-                 .contiguous();
-    auto k = qkv.narrow(1, q_dim, kv_dim)
-                 .view({T, m.n_kv_heads, m.head_dim}).contiguous();
-    auto v = qkv.narrow(1, q_dim + kv_dim, kv_dim)
-                 .view({T, m.n_kv_heads, m.head_dim}).contiguous();
+    // one fused pass instead of three narrow().contiguous() copy chains
+    split_qkv(q, k, v, qkv);
     rope_inplace(q, k, positions, m.rope_theta);
     kv_append(kcaches[li], vcaches[li], k, v, slots);
     attend(m, attn_out, q, kcaches[li], vcaches[li], block_tables, n_decode,
