@@ -18,8 +18,13 @@ def test_groves_listed():
 def test_qa_benchmark_grove_contents():
     g = G.load_grove(os.path.join(ROOT, "qa-benchmark"))
     assert g["bootstrap"]["task_description"].startswith("Run the QA")
-    # *_file bootstrap fields resolve to file contents
-    assert "blorks" in g["bootstrap"]["immediate_context"]
+    # *_file bootstrap fields resolve to file contents — the full
+    # 600-question procedurally-generated bank (scripts/gen_qa_bank.py)
+    import json as _json
+    bank = _json.loads(g["bootstrap"]["immediate_context"])
+    assert len(bank) == 6 and all(len(v) == 100 for v in bank.values())
+    one = bank["arithmetic"][0]
+    assert len(one["options"]) == 10 and one["answer"] in "ABCDEFGHIJ"
     edge = g["topology"]["edges"][0]
     assert edge["auto_inject"]["profile"] == "default"
     assert g["workspace"] == "scratch"

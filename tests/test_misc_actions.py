@@ -159,13 +159,16 @@ async def test_skills_create_learn_and_grove_shadowing(tmp_path):
 
 
 @pytest.mark.asyncio
-async def test_generate_images_without_model_is_structured():
+async def test_generate_images_without_model_uses_local_renderer():
+    """No injected image_fn: the locally-hosted procedural model renders
+    a real PNG and the artifact pipeline stores it (utils/imagegen.py)."""
     runtime = make_runtime()
     actor = _actor(runtime)
     res = await R.execute_action(_ctx(actor, runtime, "generate_images",
                                       {"prompt": "a red square"}))
     assert isinstance(res, dict)
-    assert res.get("error") or res.get("status")
+    assert res.get("model") == "local-procedural-v0"
+    assert res.get("image_artifacts")
 
 
 @pytest.mark.asyncio
