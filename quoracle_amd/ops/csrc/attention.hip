@@ -610,7 +610,9 @@ paged_attn_prefill_split_kernel(
   }
 }
 
-// Combine prefill split partials.  Grid: (ntiles, Hq); block: DECODE_BLOCK.
+// Combine prefill split partials.  Grid: (ntiles, Hq, QT); block:
+// DECODE_BLOCK.  One workgroup per query row — small grammar chunks
+// launch few (tile, head) pairs, so the row loop must not serialize.
 extern "C" __global__ void __launch_bounds__(DECODE_BLOCK)
 paged_attn_prefill_reduce_kernel(bf16 *__restrict__ out,
                                  const float *__restrict__ part_m,
@@ -621,33 +623,31 @@ paged_attn_prefill_reduce_kernel(bf16 *__restrict__ out,
                                  int Hq, int D, int NS) {
   const int tile = blockIdx.x;
   const int h = blockIdx.y;
+  const int qi = blockIdx.z;
   const int tid = threadIdx.x;
   const int q0 = tile_q0[tile];
-  const int qn = tile_qn[tile];
+  if (qi >= tile_qn[tile]) return;
   __shared__ float scratch[8];
 
-  for (int qi = 0; qi < qn; ++qi) {
-    const long base0 = (((long)tile * Hq + h) * NS) * QT + qi;
-    float M = -INFINITY;
-    for (int i = tid; i < NS; i += blockDim.x)
-      M = fmaxf(M, part_m[base0 + (long)i * QT]);
-    M = block_max(M, scratch);
-    float L = 0.f;
-    for (int i = tid; i < NS; i += blockDim.x) {
+  const long base0 = (((long)tile * Hq + h) * NS) * QT + qi;
+  float M = -INFINITY;
+  for (int i = tid; i < NS; i += blockDim.x)
+    M = fmaxf(M, part_m[base0 + (long)i * QT]);
+  M = block_max(M, scratch);
+  float L = 0.f;
+  for (int i = tid; i < NS; i += blockDim.x) {
+    float mi = part_m[base0 + (long)i * QT];
+    L += (mi == -INFINITY) ? 0.f : part_l[base0 + (long)i * QT] * __expf(mi - M);
+  }
+  L = block_sum(L, scratch);
+  if (tid < D) {
+    float o = 0.f;
+    for (int i = 0; i < NS; ++i) {
       float mi = part_m[base0 + (long)i * QT];
-      L += (mi == -INFINITY) ? 0.f : part_l[base0 + (long)i * QT] * __expf(mi - M);
+      if (mi == -INFINITY) continue;
+      o = fmaf(part_acc[(base0 + (long)i * QT) * D + tid], __expf(mi - M), o);
     }
-    L = block_sum(L, scratch);
-    if (tid < D) {
-      float o = 0.f;
-      for (int i = 0; i < NS; ++i) {
-        float mi = part_m[base0 + (long)i * QT];
-        if (mi == -INFINITY) continue;
-        o = fmaf(part_acc[(base0 + (long)i * QT) * D + tid], __expf(mi - M), o);
-      }
-      out[((long)(q0 + qi) * Hq + h) * D + tid] = f2bf(L > 0.f ? o / L : 0.f);
-    }
-    __syncthreads();
+    out[((long)(q0 + qi) * Hq + h) * D + tid] = f2bf(L > 0.f ? o / L : 0.f);
   }
 }
 

@@ -339,7 +339,7 @@ void paged_attn_prefill_split(torch::Tensor out, torch::Tensor q,
                      tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, D, BS,
                      MAXB, GQ, NS);
   HIP_CHECK_KERNEL();
-  hipLaunchKernelGGL(paged_attn_prefill_reduce_kernel, dim3(ntiles, Hq),
+  hipLaunchKernelGGL(paged_attn_prefill_reduce_kernel, dim3(ntiles, Hq, 16),
                      dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
                      part_m.data_ptr<float>(), part_l.data_ptr<float>(),
                      part_acc.data_ptr<float>(), tile_q0.data_ptr<int>(),
@@ -414,7 +414,7 @@ void paged_attn_prefill_mfma_split(
                      tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, BS,
                      MAXB, GQ, NS);
   HIP_CHECK_KERNEL();
-  hipLaunchKernelGGL(paged_attn_prefill_reduce_kernel, dim3(ntiles, Hq),
+  hipLaunchKernelGGL(paged_attn_prefill_reduce_kernel, dim3(ntiles, Hq, 16),
                      dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
                      part_m.data_ptr<float>(), part_l.data_ptr<float>(),
                      part_acc.data_ptr<float>(), tile_q0.data_ptr<int>(),
