@@ -588,7 +588,54 @@ async def execute_call_mcp(ctx) -> Dict[str, Any]:
 # answer_engine / generate_images
 # ---------------------------------------------------------------------------
 
+def _answer_corpus_dirs(ctx) -> List[str]:
+    """Grounding corpus: an explicit answer_corpus_dir plus the agent's
+    grove directory/workspace (the locally-reachable knowledge this
+    environment has in place of the web)."""
+    dirs: List[str] = []
+    extra = ctx.runtime.extras.get("answer_corpus_dir") \
+        or getattr(ctx.runtime.config, "answer_corpus_dir", None)
+    if extra:
+        dirs.append(extra)
+    grove = ctx.agent.state.grove or {}
+    gdir = grove.get("path")
+    if gdir:
+        dirs.append(gdir)
+    return [d for d in dirs if os.path.isdir(d)]
+
+
+def _gather_passages(dirs: List[str], max_files: int = 64,
+                     chunk_chars: int = 1200) -> List[Dict[str, str]]:
+    passages: List[Dict[str, str]] = []
+    exts = {".md", ".txt", ".json", ".py", ".yaml", ".yml", ".csv"}
+    for base in dirs:
+        for root, _dirnames, files in os.walk(base):
+            for name in sorted(files):
+                if os.path.splitext(name)[1].lower() not in exts:
+                    continue
+                path = os.path.join(root, name)
+                try:
+                    with open(path, "r", errors="replace") as f:
+                        text = f.read(256 * 1024)
+                except OSError:
+                    continue
+                for off in range(0, len(text), chunk_chars):
+                    passages.append({"source": path,
+                                     "text": text[off:off + chunk_chars]})
+                if len(passages) >= max_files * 4:
+                    return passages
+    return passages
+
+
 async def execute_answer_engine(ctx) -> Dict[str, Any]:
+    """Grounded Q&A.  The reference's answer engine is web-grounded
+    (reference: actions/answer_engine.ex:28); with no egress, grounding
+    here is retrieval over the locally-reachable corpus (grove dir +
+    configured answer_corpus_dir): passages are ranked by embedding
+    cosine against the question (the same GPU embed + cosine path the
+    consensus vote uses) and injected as context, with sources reported.
+    With no corpus available it degrades to an ungrounded local answer
+    (documented divergence, PARITY.md)."""
     prompt = ctx.params["prompt"]
     model = (ctx.runtime.config.model_roles.get("answer_engine")
              or ctx.runtime.extras.get("answer_engine_model")
@@ -596,15 +643,44 @@ async def execute_answer_engine(ctx) -> Dict[str, Any]:
                  if ctx.agent.state.model_pool else None))
     if model is None:
         return _err("no_answer_engine_model")
+
+    sources: List[str] = []
+    context_block = ""
+    passages = _gather_passages(_answer_corpus_dirs(ctx))
+    if passages:
+        try:
+            facade = ctx.runtime.engines.embed_facade
+            vecs = facade([prompt] + [p["text"] for p in passages])
+            from ..consensus.rules import cosine_similarity
+            scored = sorted(
+                ((cosine_similarity(vecs[0], v), p)
+                 for v, p in zip(vecs[1:], passages)),
+                key=lambda sv: -sv[0])[:4]
+            parts = []
+            for score, p in scored:
+                parts.append(f"[source: {p['source']}]\n{p['text']}")
+                if p["source"] not in sources:
+                    sources.append(p["source"])
+            context_block = ("Grounding passages (local corpus):\n\n"
+                             + "\n\n---\n\n".join(parts) + "\n\n")
+        except Exception:  # noqa: BLE001 — grounding is best-effort
+            sources = []
+            context_block = ""
+
     engine = ctx.runtime.engines.engine_for(model)
     result = await engine.generate(GenerateRequest(
         model_key=model,
         messages=[{"role": "user",
-                   "content": "Answer factually and concisely:\n" + prompt}],
+                   "content": context_block
+                   + "Answer factually and concisely:\n" + prompt}],
         temperature=0.3, max_tokens=2048))
     if not result.ok:
         return _err("answer_engine_failed", detail=result.error)
-    return {"answer": result.text, "model": model}
+    out = {"answer": result.text, "model": model,
+           "grounded": bool(sources)}
+    if sources:
+        out["sources"] = sources
+    return out
 
 
 async def execute_generate_images(ctx) -> Dict[str, Any]:

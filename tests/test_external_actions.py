@@ -199,3 +199,44 @@ async def test_call_mcp_http_roundtrip():
         assert "sess-42" in state["session_seen"]
     finally:
         srv.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_answer_engine_grounds_on_local_corpus(tmp_path):
+    """answer_engine retrieves from the local corpus (the no-egress
+    grounding analog of the reference's web-grounded engine): the
+    best-matching passage's source is reported and its content reaches
+    the model prompt."""
+    (tmp_path / "phobos.md").write_text(
+        "Phobos is the larger of the two moons of Mars. "
+        "Its orbital period is 7 hours 39 minutes.")
+    (tmp_path / "cooking.md").write_text(
+        "Sourdough needs a mature starter and long fermentation.")
+    seen = {}
+
+    def responder(model, msgs, req):
+        seen["prompt"] = msgs[-1]["content"]
+        return '{"reasoning": "r", "action": "wait", "params": {}}'
+
+    from quoracle_amd.engine.fake import FakeEngine
+    engine = FakeEngine(response_fn=responder)
+    runtime = make_runtime(engine=engine)
+    runtime.extras["answer_corpus_dir"] = str(tmp_path)
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "answer_engine", {
+        "prompt": "What is the orbital period of Phobos around Mars?"}))
+    assert res.get("grounded") is True, res
+    assert any("phobos.md" in s for s in res["sources"]), res
+    assert "7 hours 39 minutes" in seen["prompt"]
+    # untrusted wrapping still applies to the answer
+    assert "NO_EXECUTE_" in json.dumps(res)
+
+
+@pytest.mark.asyncio
+async def test_answer_engine_without_corpus_is_ungrounded():
+    runtime = make_runtime()
+    actor = _actor(runtime)
+    res = await R.execute_action(_ctx(actor, runtime, "answer_engine",
+                                      {"prompt": "anything"}))
+    assert res.get("grounded") is False
+    assert "sources" not in res
