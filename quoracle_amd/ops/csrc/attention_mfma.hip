@@ -672,8 +672,11 @@ paged_attn_prefill_mfma64_kernel(
   // row-major staging writes are conflict-free b32s, and VR = 134
   // (67 dwords, odd) spreads the PV B-fragment's scalar key-gather over
   // all banks: key groups land at +24k mod 64, d-columns at +c/2.
-  __shared__ bf16 q_s[MF4_QT * KP];
-  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  // q_s/k_s: XOR-swizzled [row][128] images (guide G4: elem block index
+  // ^ (row&15)<<3) — zero-conflict b128 A/B-fragment reads at the full
+  // 64-lane phase, where +8 padding still left 2-way phase conflicts
+  __shared__ bf16 q_s[MF4_QT * MF_D];
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];
   __shared__ bf16 v_s[MF_KCHUNK * VR];
   __shared__ float s_s[MF4_QT * SP];
   __shared__ bf16 p_s[MF4_QT * VP];
@@ -685,7 +688,8 @@ paged_attn_prefill_mfma64_kernel(
     if (r < qn)
       val = reinterpret_cast<const uint4 *>(
           q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
-    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
   }
   if (tid < MF4_QT) {
     m_s[tid] = -INFINITY;
@@ -732,7 +736,8 @@ paged_attn_prefill_mfma64_kernel(
     for (int it = 0; it < 2; ++it) {
       const int i = tid + it * 512;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
       // V row-major: 4 b32 stores (VR row stride is 4-byte aligned only);
       // conflict-free — lanes sharing a key span banks 4 apart, the four
       // keys per instruction land at distinct bank residues (VR/2 odd)
@@ -756,12 +761,13 @@ paged_attn_prefill_mfma64_kernel(
       s_acc1 = (f32x4_t){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < MF_D / 32; ++kk) {
+        const int koff = (kk * 32 + a_koff) ^ ((a_row & 15) << 3);
         bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
-            k_s + (kblock * 16 + a_row) * KP + kk * 32 + a_koff);
+            k_s + (kblock * 16 + a_row) * MF_D + koff);
         bf16x8_t a0 = *reinterpret_cast<const bf16x8_t *>(
-            q_s + ((qb0 + 0) * 16 + a_row) * KP + kk * 32 + a_koff);
+            q_s + ((qb0 + 0) * 16 + a_row) * MF_D + koff);
         bf16x8_t a1 = *reinterpret_cast<const bf16x8_t *>(
-            q_s + ((qb0 + 1) * 16 + a_row) * KP + kk * 32 + a_koff);
+            q_s + ((qb0 + 1) * 16 + a_row) * MF_D + koff);
         s_acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, s_acc0, 0, 0, 0);
         s_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, s_acc1, 0, 0, 0);
       }
