@@ -60,8 +60,9 @@ paged_attn_prefill_mfma_kernel(
   const int pos0 = tile_pos0[tile];
   const int kv_limit = pos0 + qn;
 
-  __shared__ bf16 q_s[MF_QT * KP];
-  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  // q_s/k_s: XOR-swizzled [row][128] images (elem block ^ (row&15)<<3)
+  __shared__ bf16 q_s[MF_QT * MF_D];
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];
   __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
@@ -74,7 +75,8 @@ paged_attn_prefill_mfma_kernel(
     if (r < qn)
       val = reinterpret_cast<const uint4 *>(
           q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
-    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
   }
   if (tid < MF_QT) {
     m_s[tid] = -INFINITY;
@@ -119,7 +121,8 @@ paged_attn_prefill_mfma_kernel(
     for (int it = 0; it < 4; ++it) {
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
       const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
@@ -143,9 +146,10 @@ paged_attn_prefill_mfma_kernel(
 #pragma unroll
       for (int kk = 0; kk < MF_D / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
-            q_s + a_row * KP + kk * 32 + a_koff);
+            q_s + a_row * MF_D + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
-            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+            k_s + (wave * 16 + a_row) * MF_D
+                + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
       }
 #pragma unroll
@@ -269,8 +273,9 @@ paged_attn_prefill_mfma_split_kernel(
   const int c1 = min(kv_limit, (split + 1) * per * MF_KCHUNK);
   const bool dead = (c0 >= kv_limit);
 
-  __shared__ bf16 q_s[MF_QT * KP];
-  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  // q_s/k_s: XOR-swizzled [row][128] images (elem block ^ (row&15)<<3)
+  __shared__ bf16 q_s[MF_QT * MF_D];
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];
   __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
@@ -282,7 +287,8 @@ paged_attn_prefill_mfma_split_kernel(
     if (r < qn)
       val = reinterpret_cast<const uint4 *>(
           q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
-    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
   }
   if (tid < MF_QT) {
     m_s[tid] = -INFINITY;
@@ -326,7 +332,8 @@ paged_attn_prefill_mfma_split_kernel(
     for (int it = 0; it < 4; ++it) {
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
       const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
@@ -348,9 +355,10 @@ paged_attn_prefill_mfma_split_kernel(
 #pragma unroll
       for (int kk = 0; kk < MF_D / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
-            q_s + a_row * KP + kk * 32 + a_koff);
+            q_s + a_row * MF_D + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
-            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+            k_s + (wave * 16 + a_row) * MF_D
+                + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
       }
 #pragma unroll
@@ -469,8 +477,9 @@ paged_attn_prefill_mfma32_kernel(
   const int pos0 = tile_pos0[tile];
   const int kv_limit = pos0 + qn;
 
-  __shared__ bf16 q_s[MF2_QT * KP];
-  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  // q_s/k_s: XOR-swizzled [row][128] images (elem block ^ (row&15)<<3)
+  __shared__ bf16 q_s[MF2_QT * MF_D];
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];
   __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF2_QT * SP];
   __shared__ bf16 p_s[MF2_QT * VP];
@@ -482,7 +491,8 @@ paged_attn_prefill_mfma32_kernel(
     if (r < qn)
       val = reinterpret_cast<const uint4 *>(
           q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
-    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
   }
   if (tid < MF2_QT) {
     m_s[tid] = -INFINITY;
@@ -526,7 +536,8 @@ paged_attn_prefill_mfma32_kernel(
     for (int it = 0; it < 2; ++it) {
       const int i = tid + it * 512;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
       const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
@@ -548,9 +559,11 @@ paged_attn_prefill_mfma32_kernel(
 #pragma unroll
       for (int kk = 0; kk < MF_D / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
-            q_s + (qblock * 16 + a_row) * KP + kk * 32 + a_koff);
+            q_s + (qblock * 16 + a_row) * MF_D
+                + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         bf16x8_t b = *reinterpret_cast<const bf16x8_t *>(
-            k_s + (kblock * 16 + a_row) * KP + kk * 32 + a_koff);
+            k_s + (kblock * 16 + a_row) * MF_D
+                + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, s_acc, 0, 0, 0);
       }
 #pragma unroll
@@ -906,8 +919,9 @@ paged_attn_decode_mfma_kernel(
   const int t1 = min(len, (split + 1) * per * MF_KCHUNK);
   const bool dead = (t0 >= len);
 
-  __shared__ bf16 q_s[MF_QT * KP];
-  __shared__ bf16 k_s[MF_KCHUNK * KP];
+  // q_s/k_s: XOR-swizzled [row][128] images (elem block ^ (row&15)<<3)
+  __shared__ bf16 q_s[MF_QT * MF_D];
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];
   __shared__ bf16 v_s[MF_KCHUNK * VR];   // row-major V (see VR note)
   __shared__ float s_s[MF_QT * SP];
   __shared__ bf16 p_s[MF_QT * VP];
@@ -920,7 +934,8 @@ paged_attn_decode_mfma_kernel(
     if (r < GQ)
       val = reinterpret_cast<const uint4 *>(
           q + ((long)b * Hq + hk * GQ + r) * MF_D + c)[0];
-    reinterpret_cast<uint4 *>(q_s + r * KP + c)[0] = val;
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
   }
   if (tid < MF_QT) {
     m_s[tid] = -INFINITY;
@@ -961,7 +976,8 @@ paged_attn_decode_mfma_kernel(
     for (int it = 0; it < 4; ++it) {
       const int i = tid + it * 256;
       const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
-      reinterpret_cast<uint4 *>(k_s + key * KP + d)[0] = kreg[it];
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
       const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
@@ -983,9 +999,10 @@ paged_attn_decode_mfma_kernel(
 #pragma unroll
       for (int kk = 0; kk < MF_D / 32; ++kk) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
-            q_s + a_row * KP + kk * 32 + a_koff);
+            q_s + a_row * MF_D + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         bf16x8_t b2 = *reinterpret_cast<const bf16x8_t *>(
-            k_s + (wave * 16 + a_row) * KP + kk * 32 + a_koff);
+            k_s + (wave * 16 + a_row) * MF_D
+                + ((kk * 32 + a_koff) ^ ((a_row & 15) << 3)));
         s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, s_acc, 0, 0, 0);
       }
 #pragma unroll

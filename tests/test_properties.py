@@ -442,3 +442,53 @@ def test_skill_parser_total(text):
         assert isinstance(out, dict)
     except (SkillError, ValueError, KeyError):
         pass
+
+
+@settings(max_examples=60, deadline=None)
+@given(ids=st.lists(st.integers(min_value=0, max_value=10 ** 6),
+                    min_size=0, max_size=4200),
+       subset=st.sets(st.sampled_from(
+           __import__("quoracle_amd.actions.schema",
+                      fromlist=["ACTIONS"]).ACTIONS), min_size=0, max_size=22),
+       child=st.one_of(st.none(), st.text(max_size=12)))
+def test_grammar_walk_total_over_all_actions(ids, subset, child):
+    """Any sampled-id stream through the 22-action grammar terminates in
+    bounded steps and emits parseable JSON (or the stream simply ran
+    short) — no crashes for any action subset or context."""
+    ctx = {"child_id": child} if child is not None else {}
+    g = ActionGrammar(sorted(subset) or ["wait"], context=ctx)
+    out = []
+    for sid in ids:
+        if g.done:
+            break
+        out.append(g.advance(sid))
+    if g.done:
+        text = bytes(b for b in out if b != EOS).decode()
+        doc = json.loads(text)
+        assert doc["action"] in (sorted(subset) or ["wait"])
+
+
+@settings(max_examples=80, deadline=None)
+@given(text=st.text(max_size=400),
+       want=st.integers(min_value=0, max_value=5))
+def test_mcp_sse_parser_total(text, want):
+    """SSE bodies from an HTTP MCP server are untrusted bytes: parse the
+    matching message or raise ConnectionError, never crash."""
+    from quoracle_amd.actions.executors import MCPHttpConnection
+    try:
+        msg = MCPHttpConnection._from_sse(text, want)
+        assert isinstance(msg, dict) and msg.get("id") == want
+    except ConnectionError:
+        pass
+
+
+@settings(max_examples=25, deadline=None)
+@given(prompt=st.text(max_size=200),
+       src=st.one_of(st.none(), st.binary(max_size=64)))
+def test_imagegen_always_valid_png(prompt, src):
+    """The procedural image model renders a structurally valid PNG for
+    ANY prompt/source, deterministically."""
+    from quoracle_amd.utils import imagegen
+    png = imagegen.render(prompt, size=(32, 24), source_image=src)
+    assert png.startswith(b"\x89PNG\r\n\x1a\n")
+    assert png == imagegen.render(prompt, size=(32, 24), source_image=src)
