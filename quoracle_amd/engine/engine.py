@@ -520,9 +520,14 @@ class LocalEngine(Engine):
 
         row = n_decode
         max_kv = 0
-        # 64-row tiles match the default mfma64 prefill kernel (145 TF,
-        # validated r2); QUORACLE_MFMA32 falls back to 32-row tiles
-        big_step = 32 if os.environ.get("QUORACLE_MFMA32") else 64
+        # 128-row tiles match the default T12 prefill kernel (352 TF,
+        # validated r2); QUORACLE_MFMA64/QUORACLE_MFMA32 fall back
+        if os.environ.get("QUORACLE_MFMA32"):
+            big_step = 32
+        elif os.environ.get("QUORACLE_MFMA64"):
+            big_step = 64
+        else:
+            big_step = 128
         for seq, n in prefill:
             cached = len(seq.session.token_ids)
             chunk = seq.known[cached:cached + n]

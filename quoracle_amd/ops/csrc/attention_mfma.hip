@@ -1090,3 +1090,236 @@ paged_attn_decode_mfma_kernel(
     part_acc[base * MF_D + wave * 32 + 16 + c_col] = dead ? 0.f : o_acc1[r];
   }
 }
+
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL T12 prefill: 128-row Q tiles, 8 waves, each wave OWNS one
+// 16-row qblock end to end.  Swapped QK^T (mfma(K, Q)) makes every S row
+// lane-local, so the online softmax runs fully in registers (2 shfl_xor
+// across the 4 same-qrow lanes) — no s_s/p_s LDS round trip and only TWO
+// barriers per chunk (staging).  The P->A-fragment redistribution is the
+// guide's T12 permlane dance: pack P to bf16 dwords, then per 32-key
+// k-block a pl32swap + pl16swap pair leaves every lane's A-fragment as
+// the uniform dword sequence [d0, d1, e0, e1].
+//   K: XOR-swizzled [64][128] image (shared staging, as mfma64)
+//   V: row-major [64][VR] image; PV B-fragments by conflict-free scalar
+//      key-gather (as mfma64)
+//   alpha/l cross-lane handoff (softmax lanes hold qrow=lane&15; PV C
+//   rows live at (lane>>4)*4+reg) via a tiny per-wave LDS strip.
+// Grid: (ntiles128, Hq); block 512.
+// ---------------------------------------------------------------------------
+#define T12_QT 128
+
+extern "C" __global__ void __launch_bounds__(512, 4)
+paged_attn_prefill_t12_kernel(
+    bf16 *__restrict__ out, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];     // XOR-swizzled
+  __shared__ bf16 v_s[MF_KCHUNK * VR];       // row-major
+  __shared__ float alpha_w[8][16];
+  __shared__ float l_w[8][16];
+
+  const int qr = lane & 15;                  // this lane's qrow (in qblock)
+  const int g = lane >> 4;                   // 16-lane group
+  const int row_local = wave * 16 + qr;      // row within the 128-row tile
+
+  // Q B-fragments in registers: lane holds Q[qrow][k = g*8 + t] per kk
+  bf16x8_t q_frag[MF_D / 32];
+#pragma unroll
+  for (int kk = 0; kk < MF_D / 32; ++kk) {
+    bf16x8_t v = {};
+    if (row_local < qn)
+      v = *reinterpret_cast<const bf16x8_t *>(
+          q + ((long)(q0 + row_local) * Hq + h) * MF_D + kk * 32 + g * 8);
+    q_frag[kk] = v;
+  }
+
+  // per-lane online-softmax state for qrow (replicated over the 4
+  // same-qrow lanes, deterministically identical)
+  float m_run = -INFINITY, l_run = 0.f;
+  // O accumulators: 8 d-blocks x f32x4 (C rows = qrows (g*4+r), col = d)
+  f32x4_t o_acc[8];
+#pragma unroll
+  for (int db = 0; db < 8; ++db) o_acc[db] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int abs_qrow = min(row_local, qn - 1) + pos0;   // causal bound
+
+  uint4 kreg[2], vreg[2];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int i = tid + it * 512;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
+    }
+  };
+
+  issue_loads(0, kv_limit);
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+    write_staged();
+    __syncthreads();
+    if (start + MF_KCHUNK < kv_limit)
+      issue_loads(start + MF_KCHUNK, kv_limit);
+
+    // ---- S^T = K·Q^T per 16-key block: C[key][qrow], lane-local rows --
+    f32x4_t s_frag[4];
+#pragma unroll
+    for (int kb = 0; kb < 4; ++kb) {
+      f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < MF_D / 32; ++kk) {
+        // A = K rows (keys kb*16 + qr), k-slice g*8 (+kk*32), swizzled
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (kb * 16 + qr) * MF_D
+            + ((kk * 32 + g * 8) ^ ((qr & 15) << 3)));
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, q_frag[kk], acc,
+                                                      0, 0, 0);
+      }
+      s_frag[kb] = acc;
+    }
+
+    // ---- in-register online softmax over the 64-key chunk -----------
+    // (masked scores, then P, overwrite s_frag in place — the S pipeline
+    // must not triple register pressure: vals/pv/s_frag were 48 VGPRs)
+    float mymax = -INFINITY;
+#pragma unroll
+    for (int kb = 0; kb < 4; ++kb) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = kb * 16 + g * 4 + r;
+        const int token = start + key;
+        const bool ok = (token <= abs_qrow) && (key < clen);
+        const float sv = ok ? s_frag[kb][r] * scale : -INFINITY;
+        s_frag[kb][r] = sv;
+        mymax = fmaxf(mymax, sv);
+      }
+    }
+    mymax = fmaxf(mymax, __shfl_xor(mymax, 16));
+    mymax = fmaxf(mymax, __shfl_xor(mymax, 32));
+    // (T13 defer-max measured perf-neutral here at 3x the rounding error
+    // — the rescale pass is only 32 VALU ops — so the exact path stays)
+    const float mn = fmaxf(m_run, mymax);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - mn);
+    float psum = 0.f;
+#pragma unroll
+    for (int kb = 0; kb < 4; ++kb) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float p = (s_frag[kb][r] == -INFINITY || mn == -INFINITY)
+                            ? 0.f : __expf(s_frag[kb][r] - mn);
+        psum += p;
+        s_frag[kb][r] = p;
+      }
+    }
+    psum += __shfl_xor(psum, 16);
+    psum += __shfl_xor(psum, 32);
+    l_run = l_run * alpha + psum;
+    m_run = mn;
+    if (lane < 16) alpha_w[wave][lane] = alpha;   // qrow == lane here
+
+    // ---- O rescale (alpha for PV C rows g*4+r via the wave strip) ----
+#pragma unroll
+    for (int db = 0; db < 8; ++db) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[db][r] *= alpha_w[wave][g * 4 + r];
+    }
+
+    // ---- P -> A fragments (permlane dance) + PV ----------------------
+    auto packbf = [](float lo, float hi) {
+      union { bf16 h; unsigned short u; } a, b;
+      a.h = f2bf(lo);
+      b.h = f2bf(hi);
+      return (uint)a.u | ((uint)b.u << 16);
+    };
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      // pack this k-block's two key-blocks (kb0 = 2kk, kb1 = 2kk+1)
+      uint d0 = packbf(s_frag[2 * kk][0], s_frag[2 * kk][1]);
+      uint d1 = packbf(s_frag[2 * kk][2], s_frag[2 * kk][3]);
+      uint e0 = packbf(s_frag[2 * kk + 1][0], s_frag[2 * kk + 1][1]);
+      uint e1 = packbf(s_frag[2 * kk + 1][2], s_frag[2 * kk + 1][3]);
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(d0, e0, false, false);
+        d0 = r[0]; e0 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(d1, e1, false, false);
+        d1 = r[0]; e1 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane16_swap(d0, e0, false, false);
+        d0 = r[0]; e0 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane16_swap(d1, e1, false, false);
+        d1 = r[0]; e1 = r[1];
+      }
+      const uint4 av = make_uint4(d0, d1, e0, e1);
+      const bf16x8_t a_frag = __builtin_bit_cast(bf16x8_t, av);
+#pragma unroll
+      for (int db = 0; db < 8; ++db) {
+        bf16x8_t b;
+#pragma unroll
+        for (int t = 0; t < 8; ++t)
+          b[t] = v_s[(kk * 32 + g * 8 + t) * VR + db * 16 + qr];
+        o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b,
+                                                            o_acc[db], 0, 0, 0);
+      }
+    }
+    __syncthreads();   // next chunk restages k_s / v_s
+  }
+
+  if (lane < 16) l_w[wave][lane] = l_run;
+  // epilogue: PV C rows = qrows g*4+r of this wave's qblock
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = wave * 16 + g * 4 + r;
+    if (row >= qn) continue;
+    const float lv = l_w[wave][g * 4 + r];
+    const float denom = lv > 0.f ? lv : 1.f;
+#pragma unroll
+    for (int db = 0; db < 8; ++db)
+      out[((long)(q0 + row) * Hq + h) * MF_D + db * 16 + qr] =
+          f2bf(o_acc[db][r] / denom);
+  }
+}

@@ -480,6 +480,37 @@ void paged_attn_prefill_mfma32(torch::Tensor out, torch::Tensor q,
   HIP_CHECK_KERNEL();
 }
 
+void paged_attn_prefill_t12(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor kcache, torch::Tensor vcache,
+                            torch::Tensor block_tables,
+                            torch::Tensor tile_q0, torch::Tensor tile_qn,
+                            torch::Tensor tile_seq,
+                            torch::Tensor tile_pos0, double scale) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  const int ntiles = tile_q0.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  TORCH_CHECK(D == 128, "T12 prefill kernel requires head dim 128");
+  TORCH_CHECK(Hq % Hkv == 0, "bad GQA ratio");
+  if (ntiles == 0) return;
+  hipLaunchKernelGGL(paged_attn_prefill_t12_kernel, dim3(ntiles, Hq),
+                     dim3(512), 0, current_stream(), bf16_ptr(out),
+                     bf16_cptr(q), bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                     tile_qn.data_ptr<int>(), tile_seq.data_ptr<int>(),
+                     tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, BS,
+                     MAXB, GQ);
+  HIP_CHECK_KERNEL();
+}
+
 void paged_attn_prefill_mfma64(torch::Tensor out, torch::Tensor q,
                                torch::Tensor kcache, torch::Tensor vcache,
                                torch::Tensor block_tables,
@@ -538,6 +569,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Context-split prefill attention + combine (small chunks)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
         "MFMA-tiled prefill attention (D=128, matrix cores)");
+  m.def("paged_attn_prefill_t12", &paged_attn_prefill_t12,
+        "EXPERIMENTAL: 128-row T12 prefill (swapped QK^T, in-register "
+        "softmax, permlane P exchange; validate before use)");
   m.def("paged_attn_prefill_mfma64", &paged_attn_prefill_mfma64,
         "EXPERIMENTAL: MFMA prefill, 64-row Q tiles (validate before use)");
   m.def("paged_attn_prefill_mfma32", &paged_attn_prefill_mfma32,

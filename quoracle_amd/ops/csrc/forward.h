@@ -131,19 +131,24 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
                                       *tile_seq, *tile_pos0, m.scale, pm, pl,
                                       pa);
       } else if (t32_q0.has_value() && t32_q0->numel() > 0) {
-        // big prefill: 8-wave big tiles (2-4x K/V reuse per query row).
-        // 64-row tiles are the default (145 TF vs 83 TF for 32-row,
-        // validated r2); QUORACLE_MFMA32 falls back (the engine sizes the
-        // tiles to match)
-        static const bool use64 = std::getenv("QUORACLE_MFMA32") == nullptr;
-        if (use64)
+        // big prefill: T12 128-row tiles are the default (352 TF measured
+        // vs 267 for mfma64 / 169 for mfma32 on the bench shape, r2);
+        // QUORACLE_MFMA64 / QUORACLE_MFMA32 fall back (the engine sizes
+        // the tiles to match)
+        static const bool use32 = std::getenv("QUORACLE_MFMA32") != nullptr;
+        static const bool use64 = std::getenv("QUORACLE_MFMA64") != nullptr;
+        if (use32)
+          paged_attn_prefill_mfma32(attn_out, q, kcache, vcache,
+                                    block_tables, *t32_q0, *t32_qn,
+                                    *t32_seq, *t32_pos0, m.scale);
+        else if (use64)
           paged_attn_prefill_mfma64(attn_out, q, kcache, vcache,
                                     block_tables, *t32_q0, *t32_qn,
                                     *t32_seq, *t32_pos0, m.scale);
         else
-          paged_attn_prefill_mfma32(attn_out, q, kcache, vcache,
-                                    block_tables, *t32_q0, *t32_qn,
-                                    *t32_seq, *t32_pos0, m.scale);
+          paged_attn_prefill_t12(attn_out, q, kcache, vcache,
+                                 block_tables, *t32_q0, *t32_qn,
+                                 *t32_seq, *t32_pos0, m.scale);
       } else {
         paged_attn_prefill_mfma(attn_out, q, kcache, vcache, block_tables,
                                 *tile_q0, *tile_qn, *tile_seq, *tile_pos0,

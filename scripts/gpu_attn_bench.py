@@ -108,3 +108,14 @@ if hasattr(ops.ext(), "paged_attn_decode_mfma"):
             pa[:B1].contiguous()), n=50)
         kb1 = 2 * B1 * ctx_len * Hkv * D * 2
         print(f"decode_mfma B={B1}: {m1*1e3:.1f} us ({kb1/m1/1e9:.2f} TB/s)")
+if hasattr(ops.ext(), "paged_attn_prefill_t12"):
+    nt128 = (new + 127) // 128
+    t128 = torch.arange(nt128, dtype=torch.int32, device=dev) * 128
+    qn128 = torch.clamp(torch.full_like(t128, new) - t128, max=128)
+    ts128 = torch.zeros_like(t128)
+    tp128 = t128 + cached
+    ms_t12 = timeit(lambda: ops.ext().paged_attn_prefill_t12(
+        out, q, kcache, vcache, tables, t128, qn128, ts128, tp128, scale))
+    rel12 = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
+    print(f"T12 prefill (EXPERIMENTAL): {ms_t12:.3f} ms "
+          f"({flops/ms_t12/1e9:.1f} TFLOP/s)  rel-vs-16 {rel12:.4f}")
