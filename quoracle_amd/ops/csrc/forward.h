@@ -118,7 +118,7 @@ inline void attend(const Model &m, torch::Tensor attn_out, torch::Tensor q,
   if (tile_q0.has_value() && tile_q0->numel() > 0) {
     const int64_t ntiles = tile_q0->size(0), Hq = m.n_heads;
     int64_t ns = std::max<int64_t>(
-        1, std::min<int64_t>(16, 2048 / std::max<int64_t>(1, ntiles * Hq)));
+        1, std::min<int64_t>(48, 4096 / std::max<int64_t>(1, ntiles * Hq)));
     auto opts = torch::TensorOptions()
                     .dtype(torch::kFloat32).device(q.device());
     if (D == 128 && !no_mfma) {

@@ -100,7 +100,7 @@ def paged_attn_prefill(out, q, kcache, vcache, block_tables, tile_q0,
         D = q.shape[2]
         # D=128 prefill goes to the matrix cores; small grids additionally
         # context-split so the KV walk parallelizes across the chip
-        ns = max(1, min(16, 2048 // max(1, ntiles * Hq)))
+        ns = max(1, min(48, 4096 // max(1, ntiles * Hq)))
         if D == 128 and not os.environ.get("QUORACLE_NO_MFMA_ATTN"):
             if ns > 1 and max_kv >= 1024:
                 QT = 16

@@ -221,6 +221,24 @@ def create_app(manager) -> FastAPI:
             force_reflection=body.force_reflection))
         return {"ok": True}
 
+    @app.post("/api/admin/reload")
+    def admin_reload():
+        """Hot-reload analog of the reference's `mix quoracle.reload` /
+        `llm_db.hot_reload` dev tasks: groves, skills and profiles are
+        already read from disk/DB per use, so the only cached state is
+        each live agent's system prompt — invalidate it so prompt-file /
+        grove / skill edits take effect on the next consensus cycle."""
+        invalidated = 0
+        for agent_id in runtime.registry.all_ids():
+            entry = runtime.registry.lookup(agent_id)
+            if entry is not None:
+                entry.actor.state.cached_system_prompt = None
+                invalidated += 1
+        return {"ok": True, "agents_invalidated": invalidated,
+                "hot_sources": ["groves (per-use disk read)",
+                                "skills (per-use disk read)",
+                                "profiles (DB-backed)"]}
+
     @app.get("/api/secrets")
     def secrets():
         return {"names": runtime.vault.names()}

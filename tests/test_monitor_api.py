@@ -157,3 +157,12 @@ def test_task_export_endpoint(client):
     assert isinstance(out["agents"][0]["costs"], (dict, list))
     assert isinstance(out["messages"], list)
     assert c.get("/api/tasks/t-ghost/export").status_code == 404
+
+
+def test_admin_reload_invalidates_prompt_caches(client):
+    c = client[0] if isinstance(client, tuple) else client
+    c.post("/api/tasks", json={"prompt": "reload me",
+                               "profile": "default"}).json()
+    res = c.post("/api/admin/reload").json()
+    assert res["ok"] and res["agents_invalidated"] >= 1
+    assert any("groves" in s for s in res["hot_sources"])
