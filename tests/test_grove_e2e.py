@@ -20,9 +20,15 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 @pytest.mark.asyncio
 async def test_qa_benchmark_grove_full_cycle(tmp_path):
-    # work on a COPY of the shipped grove so writes stay test-local
+    # work on a COPY of the shipped grove so writes stay test-local;
+    # shrink the 600-question bank so bootstrap immediate_context stays
+    # small (the full bank slows scripted cycles enough to flake timing)
     grove_dir = str(tmp_path / "qa-benchmark")
     shutil.copytree(os.path.join(REPO, "groves", "qa-benchmark"), grove_dir)
+    bank = json.loads(open(os.path.join(grove_dir, "questions.json")).read())
+    small = {s: qs[:2] for s, qs in list(bank.items())[:2]}
+    with open(os.path.join(grove_dir, "questions.json"), "w") as f:
+        json.dump(small, f)
     grove = G.load_grove(grove_dir)
 
     engine = FakeEngine(default_response=IDLE)
@@ -85,8 +91,12 @@ async def test_qa_benchmark_grove_full_cycle(tmp_path):
     bad = action_json("file_write", {
         "path": results_path, "mode": "write",
         "content": json.dumps({"wrong_shape": 1})})
+    # push several copies per model: if one model's queue ran a cycle
+    # ahead, round-1 unanimity fails and the refinement round re-queries
+    # — the duplicates keep every round unanimous on `bad`
     for m in POOL2:
-        engine.push_response(m, bad)
+        for _ in range(4):
+            engine.push_response(m, bad)
     await manager.send_user_message(result["task_id"], "write bad file")
     root = runtime.registry.lookup(root_id).actor
 
@@ -94,7 +104,29 @@ async def test_qa_benchmark_grove_full_cycle(tmp_path):
         h = root.state.model_histories[POOL2[0]]
         return any("schema" in str(e.get("content", "")).lower()
                    for e in h if e["type"] == "result")
-    assert await wait_until(saw_schema_error, timeout=10)
+    assert await wait_until(saw_schema_error, timeout=20)
     data = json.loads(open(results_path).read())
     assert "subjects" in data          # original file untouched
     await manager.supervisor.terminate_tree(root_id)
+
+
+@pytest.mark.asyncio
+async def test_qa_benchmark_runner_scores_bank_sample():
+    """The mmlu-pro-analog runner (scripts/run_qa_benchmark.py) drives
+    real spawn/dismiss + schema-validated confined writes over a bank
+    sample and scores exactly: oracle 1.0, noisy below it."""
+    import argparse
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "qa_runner", os.path.join(REPO, "scripts", "run_qa_benchmark.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    out = await mod.run(argparse.Namespace(
+        subjects=2, per_subject=6, engine="oracle", noise=0.0))
+    assert out["overall_accuracy"] == 1.0 and out["questions"] == 12
+    assert os.path.exists(out["results_file"])
+    data = json.loads(open(out["results_file"]).read())
+    assert set(data["subjects"]) == set(out["subjects"])
+    out2 = await mod.run(argparse.Namespace(
+        subjects=2, per_subject=12, engine="noisy", noise=0.5))
+    assert out2["overall_accuracy"] < 1.0
