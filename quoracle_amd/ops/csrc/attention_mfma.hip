@@ -38,6 +38,7 @@
                              // the PV key-gather across all LDS banks)
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -1321,5 +1322,196 @@ paged_attn_prefill_t12_kernel(
     for (int db = 0; db < 8; ++db)
       out[((long)(q0 + row) * Hq + h) * MF_D + db * 16 + qr] =
           f2bf(o_acc[db][r] / denom);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL T12-split prefill: small chunks (grammar forced-runs) over
+// long cached context.  One 16-row Q tile per workgroup, 8 waves EACH
+// walking their own interleaved 16-key chunks with WAVE-LOCAL staging (no
+// barriers in the main loop at all) and fully in-register softmax
+// (swapped QK^T).  P·V uses the 16-key mfma_f32_16x16x16bf16_1k, whose
+// A-fragment k-slice (4 keys per lane) is exactly the lane's own S rows —
+// no permlane exchange needed.  Every wave emits its own partial, so the
+// host treats waves as 8x more splits: partial layout
+//   part_m/part_l: [ntiles, Hq, NS*8, 16];  part_acc: [..., D]
+// and the existing paged_attn_prefill_reduce_kernel combines them.
+// Grid: (ntiles, Hq, NS); block 512.
+// ---------------------------------------------------------------------------
+#define TS_CHUNK 16
+
+extern "C" __global__ void __launch_bounds__(512, 4)
+paged_attn_prefill_t12_split_kernel(
+    float *__restrict__ part_m, float *__restrict__ part_l,
+    float *__restrict__ part_acc, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ, int NS) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int split = blockIdx.z;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+  const int chunks = (kv_limit + TS_CHUNK - 1) / TS_CHUNK;
+  const int per = (chunks + NS - 1) / NS;
+  const int c0 = split * per;                      // this split's chunk range
+  const int c1 = min(chunks, (split + 1) * per);
+
+  // wave-local staging buffers: no cross-wave sharing, no loop barriers
+  __shared__ bf16 k_w[8][TS_CHUNK * MF_D];         // XOR-swizzled rows
+  __shared__ bf16 v_w[8][TS_CHUNK * VR];           // row-major
+  __shared__ float alpha_ws[8][16];                // qrow alpha handoff
+
+  const int qr = lane & 15;
+  const int g = lane >> 4;
+
+  bf16x8_t q_frag[MF_D / 32];
+#pragma unroll
+  for (int kk = 0; kk < MF_D / 32; ++kk) {
+    bf16x8_t v = {};
+    if (qr < qn)
+      v = *reinterpret_cast<const bf16x8_t *>(
+          q + ((long)(q0 + qr) * Hq + h) * MF_D + kk * 32 + g * 8);
+    q_frag[kk] = v;
+  }
+
+  float m_run = -INFINITY, l_run = 0.f;
+  f32x4_t o_acc[8];
+#pragma unroll
+  for (int db = 0; db < 8; ++db) o_acc[db] = (f32x4_t){0.f, 0.f, 0.f, 0.f};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int abs_qrow = min(qr, qn - 1) + pos0;
+  bf16 *ks = k_w[wave];
+  bf16 *vs = v_w[wave];
+
+  // T14 pipelining, wave-local: chunk c+8's 16 keys ride in registers
+  // (4 x uint4 K + V per lane) while chunk c computes
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int c) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = lane + it * 64;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = c * TS_CHUNK + key;
+      if (c < c1 && token < kv_limit) {
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = lane + it * 64;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(
+          ks + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(vs + key * VR + d)[j] = vw[j];
+    }
+  };
+
+  issue_loads(c0 + wave);
+  for (int c = c0 + wave; c < c1; c += 8) {
+    const int start = c * TS_CHUNK;
+    const int clen = min(TS_CHUNK, kv_limit - start);
+    write_staged();
+    issue_loads(c + 8);
+    // wave-local LDS write->read: the compiler's lgkmcnt wait suffices
+
+    // S^T = K·Q^T (one 16-key block)
+    f32x4_t s_acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < MF_D / 32; ++kk) {
+      bf16x8_t a = *reinterpret_cast<const bf16x8_t *>(
+          ks + qr * MF_D + ((kk * 32 + g * 8) ^ ((qr & 15) << 3)));
+      s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, q_frag[kk], s_acc,
+                                                      0, 0, 0);
+    }
+
+    // in-register online softmax over the 16-key chunk
+    float mymax = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = g * 4 + r;
+      const int token = start + key;
+      const bool ok = (token <= abs_qrow) && (key < clen);
+      const float sv = ok ? s_acc[r] * scale : -INFINITY;
+      s_acc[r] = sv;
+      mymax = fmaxf(mymax, sv);
+    }
+    mymax = fmaxf(mymax, __shfl_xor(mymax, 16));
+    mymax = fmaxf(mymax, __shfl_xor(mymax, 32));
+    const float mn = fmaxf(m_run, mymax);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - mn);
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float p = (s_acc[r] == -INFINITY || mn == -INFINITY)
+                          ? 0.f : __expf(s_acc[r] - mn);
+      psum += p;
+      s_acc[r] = p;
+    }
+    psum += __shfl_xor(psum, 16);
+    psum += __shfl_xor(psum, 32);
+    l_run = l_run * alpha + psum;
+    m_run = mn;
+    // softmax state lives at qrow = lane&15; the PV C rows are qrows
+    // g*4+r — hand the alphas across lanes through the wave strip
+    if (lane < 16) alpha_ws[wave][lane] = alpha;
+
+    // A-fragment for the 16-key PV MFMA: exactly this lane's 4 P values
+    bf16x4_t a16;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) a16[r] = f2bf(s_acc[r]);
+
+#pragma unroll
+    for (int db = 0; db < 8; ++db) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[db][r] *= alpha_ws[wave][g * 4 + r];
+      bf16x4_t b;
+#pragma unroll
+      for (int t = 0; t < 4; ++t)
+        b[t] = vs[(g * 4 + t) * VR + db * 16 + qr];
+      o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(a16, b,
+                                                            o_acc[db], 0, 0, 0);
+    }
+  }
+
+  // per-wave partials at split index (split*8 + wave)
+  const bool dead = (c0 + wave >= c1);
+  const long base = (((long)tile * Hq + h) * (NS * 8) + split * 8 + wave)
+                    * QT;
+  // m/l per qrow live at lanes 0-15 (their own qrow), duplicated on the
+  // other lane groups — lane<16 writes them
+  if (lane < 16) {
+    part_m[base + lane] = dead ? -INFINITY : m_run;
+    part_l[base + lane] = dead ? 0.f : l_run;
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = g * 4 + r;             // PV C row = qrow
+#pragma unroll
+    for (int db = 0; db < 8; ++db)
+      part_acc[(base + row) * MF_D + db * 16 + qr] =
+          dead ? 0.f : o_acc[db][r];
   }
 }

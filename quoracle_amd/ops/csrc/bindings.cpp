@@ -377,6 +377,50 @@ void paged_attn_prefill_mfma(torch::Tensor out, torch::Tensor q,
   HIP_CHECK_KERNEL();
 }
 
+void paged_attn_prefill_t12_split(
+    torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
+    torch::Tensor vcache, torch::Tensor block_tables, torch::Tensor tile_q0,
+    torch::Tensor tile_qn, torch::Tensor tile_seq, torch::Tensor tile_pos0,
+    double scale, torch::Tensor part_m, torch::Tensor part_l,
+    torch::Tensor part_acc) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  const int ntiles = tile_q0.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  const int NS8 = part_m.size(2);      // waves count as splits: NS8 = NS*8
+  TORCH_CHECK(D == 128, "T12 split kernel requires head dim 128");
+  TORCH_CHECK(Hq % Hkv == 0, "bad GQA ratio");
+  TORCH_CHECK(NS8 % 8 == 0, "partial split dim must be a multiple of 8");
+  TORCH_CHECK(part_m.size(0) >= ntiles && part_m.size(1) == Hq &&
+              part_m.size(3) == QT, "workspace shape");
+  if (ntiles == 0) return;
+  hipLaunchKernelGGL(paged_attn_prefill_t12_split_kernel,
+                     dim3(ntiles, Hq, NS8 / 8), dim3(512), 0,
+                     current_stream(),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), bf16_cptr(q),
+                     bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                     tile_qn.data_ptr<int>(), tile_seq.data_ptr<int>(),
+                     tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, BS,
+                     MAXB, GQ, NS8 / 8);
+  HIP_CHECK_KERNEL();
+  hipLaunchKernelGGL(paged_attn_prefill_reduce_kernel, dim3(ntiles, Hq, 16),
+                     dim3(DECODE_BLOCK), 0, current_stream(), bf16_ptr(out),
+                     part_m.data_ptr<float>(), part_l.data_ptr<float>(),
+                     part_acc.data_ptr<float>(), tile_q0.data_ptr<int>(),
+                     tile_qn.data_ptr<int>(), Hq, D, NS8);
+  HIP_CHECK_KERNEL();
+}
+
 void paged_attn_prefill_mfma_split(
     torch::Tensor out, torch::Tensor q, torch::Tensor kcache,
     torch::Tensor vcache, torch::Tensor block_tables, torch::Tensor tile_q0,
@@ -569,6 +613,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Context-split prefill attention + combine (small chunks)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
         "MFMA-tiled prefill attention (D=128, matrix cores)");
+  m.def("paged_attn_prefill_t12_split", &paged_attn_prefill_t12_split,
+        "EXPERIMENTAL: context-split T12 prefill, wave-local chunks "
+        "(partials over NS*8 splits) + combine");
   m.def("paged_attn_prefill_t12", &paged_attn_prefill_t12,
         "EXPERIMENTAL: 128-row T12 prefill (swapped QK^T, in-register "
         "softmax, permlane P exchange; validate before use)");

@@ -119,3 +119,35 @@ if hasattr(ops.ext(), "paged_attn_prefill_t12"):
     rel12 = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
     print(f"T12 prefill (EXPERIMENTAL): {ms_t12:.3f} ms "
           f"({flops/ms_t12/1e9:.1f} TFLOP/s)  rel-vs-16 {rel12:.4f}")
+# --- small-chunk split path (grammar forced-runs: 23 new tokens over a
+# long cached context) ------------------------------------------------------
+sc_new = 23
+qsc = torch.randn(sc_new, Hq, D, device=dev, dtype=torch.bfloat16)
+osc = torch.empty_like(qsc)
+nt_sc = (sc_new + 15) // 16
+t0s = torch.arange(nt_sc, dtype=torch.int32, device=dev) * 16
+qns = torch.clamp(torch.full_like(t0s, sc_new) - t0s, max=16)
+tss = torch.zeros_like(t0s)
+tps = t0s + cached
+NSs = 16
+pms = torch.empty((nt_sc, Hq, NSs, 16), dtype=torch.float32, device=dev)
+pls = torch.empty_like(pms)
+pas = torch.empty((nt_sc, Hq, NSs, 16, D), dtype=torch.float32, device=dev)
+ms_spl = timeit(lambda: ops.ext().paged_attn_prefill_mfma_split(
+    osc, qsc, kcache, vcache, tables, t0s, qns, tss, tps, scale,
+    pms, pls, pas), n=50)
+o_spl = osc.clone()
+print(f"mfma_split small-chunk ({sc_new} tok / {cached} ctx, NS={NSs}): "
+      f"{ms_spl*1e3:.1f} us")
+if hasattr(ops.ext(), "paged_attn_prefill_t12_split"):
+    NS8 = NSs * 8
+    pm8 = torch.empty((nt_sc, Hq, NS8, 16), dtype=torch.float32, device=dev)
+    pl8 = torch.empty_like(pm8)
+    pa8 = torch.empty((nt_sc, Hq, NS8, 16, D), dtype=torch.float32,
+                      device=dev)
+    ms_t12s = timeit(lambda: ops.ext().paged_attn_prefill_t12_split(
+        osc, qsc, kcache, vcache, tables, t0s, qns, tss, tps, scale,
+        pm8, pl8, pa8), n=50)
+    rel_s = (osc.float() - o_spl.float()).norm() / o_spl.float().norm()
+    print(f"t12_split small-chunk (EXPERIMENTAL): {ms_t12s*1e3:.1f} us "
+          f"({ms_spl/ms_t12s:.2f}x)  rel-vs-mfma_split {rel_s:.4f}")
