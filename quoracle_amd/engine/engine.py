@@ -29,6 +29,7 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Tuple
 
+import numpy as np
 import torch
 
 from ..models import LlamaModel, get_config
@@ -560,9 +561,12 @@ class LocalEngine(Engine):
             return None
 
         maxb = max(len(r) for r in bt_rows)
-        bt = torch.zeros((len(bt_rows), maxb), dtype=torch.int32)
+        # numpy staging: per-row torch.tensor() creation measured ~20us
+        # each on the eager step path
+        bt_np = np.zeros((len(bt_rows), maxb), dtype=np.int32)
         for i, r in enumerate(bt_rows):
-            bt[i, :len(r)] = torch.tensor(r, dtype=torch.int32)
+            bt_np[i, :len(r)] = r
+        bt = torch.from_numpy(bt_np)
 
         def t32(x):
             return torch.tensor(x, dtype=torch.int32, device=dev)
