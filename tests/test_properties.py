@@ -492,3 +492,83 @@ def test_imagegen_always_valid_png(prompt, src):
     png = imagegen.render(prompt, size=(32, 24), source_image=src)
     assert png.startswith(b"\x89PNG\r\n\x1a\n")
     assert png == imagegen.render(prompt, size=(32, 24), source_image=src)
+
+
+@settings(max_examples=40, deadline=None)
+@given(histories=st.dictionaries(
+    st.sampled_from(["m1", "m2", "m3"]),
+    st.lists(st.one_of(
+        st.fixed_dictionaries({"type": st.sampled_from(
+            ["user", "decision", "result", "junk"]),
+            "content": st.one_of(st.text(max_size=40), st.integers(),
+                                 st.none(),
+                                 st.dictionaries(st.text(max_size=5),
+                                                 st.integers(), max_size=2))}),
+        st.dictionaries(st.text(max_size=6), st.integers(), max_size=2)),
+        max_size=6),
+    max_size=3),
+    new_pool=st.lists(st.sampled_from(["m2", "m3", "m4", "m5"]),
+                      min_size=1, max_size=3, unique=True))
+def test_history_transfer_total_over_corrupt_state(histories, new_pool):
+    """Runtime pool switching over arbitrary corrupt per-model history
+    shapes: never crashes, always leaves exactly the new pool's maps."""
+    import asyncio as _asyncio
+    from quoracle_amd.agent.history_transfer import transfer_histories
+    from quoracle_amd.agent.state import AgentState
+    from quoracle_amd.engine.fake import FakeEngine
+    state = AgentState(agent_id="f", task_id="t",
+                       model_pool=sorted(histories) or ["m1"])
+    state.init_model_maps()
+    for m, h in histories.items():
+        state.model_histories[m] = h
+    eng = FakeEngine(default_response='{"action": "wait", "params": {}}')
+    report = _asyncio.new_event_loop().run_until_complete(
+        transfer_histories(state, new_pool, lambda m: eng))
+    assert set(state.model_histories) == set(new_pool)
+    assert set(state.model_pool) == set(new_pool)
+    assert set(report) == set(new_pool)
+
+
+@settings(max_examples=50, deadline=None)
+@given(existing=st.lists(st.one_of(
+    st.fixed_dictionaries({"text": st.text(max_size=30),
+                           "confidence": st.integers(-5, 500)}),
+    st.dictionaries(st.text(max_size=5), st.integers(), max_size=2)),
+    max_size=8),
+    new=st.lists(st.one_of(
+        st.fixed_dictionaries({"text": st.text(max_size=30)}),
+        st.fixed_dictionaries({"confidence": st.integers()}),
+        st.dictionaries(st.text(max_size=4), st.text(max_size=4),
+                        max_size=2)), max_size=8))
+def test_lesson_merge_total_over_junk(existing, new):
+    """Lesson merge over malformed lesson dicts (missing text/confidence,
+    junk keys): bounded output, never crashes."""
+    from quoracle_amd.agent.lessons import MAX_LESSONS, merge_lessons
+    out = merge_lessons(existing, new)
+    assert isinstance(out, list) and len(out) <= MAX_LESSONS
+
+
+@settings(max_examples=25, deadline=None)
+@given(path_bit=st.text(
+    alphabet=st.characters(blacklist_characters="/\x00",
+                           blacklist_categories=("Cs",)),
+    min_size=1, max_size=24))
+def test_ui_paths_never_500(path_bit):
+    """Arbitrary ids in UI paths produce 2xx/4xx, never a server error
+    (the ids reach SQL parameters and registry lookups)."""
+    from fastapi.testclient import TestClient
+    from quoracle_amd.engine.fake import FakeEngine
+    from quoracle_amd.ui.server import create_app
+    from helpers import IDLE, make_manager
+    engine = FakeEngine(default_response=IDLE)
+    manager, runtime = make_manager(engine)
+    app = create_app(manager)
+    from urllib.parse import quote
+    pb = quote(path_bit, safe="")
+    with TestClient(app, raise_server_exceptions=False) as client:
+        for url in (f"/api/tasks/{pb}/tree",
+                    f"/api/agents/{pb}/logs",
+                    f"/api/agents/{pb}/state",
+                    f"/api/tasks/{pb}/export"):
+            r = client.get(url)
+            assert r.status_code < 500, (url, r.status_code, r.text[:200])
