@@ -206,6 +206,8 @@ class LocalEngine(Engine):
                     raise TimeoutError("generate_sync timed out")
         else:
             ev.wait(timeout)
+        if not box:
+            raise TimeoutError("generate_sync timed out")
         return box[0]
 
     async def embed(self, texts: List[str]) -> Sequence[Sequence[float]]:

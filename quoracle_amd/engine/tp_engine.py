@@ -56,8 +56,11 @@ class TPEngine(LocalEngine):
         # lockstep: batches are composed ONLY from the admit-broadcast —
         # requests landing between broadcast and admit wait for the next
         # step (a leader admitting unannounced work would desync the
-        # per-layer all-reduce counts and deadlock the group)
-        self._tp_direct = True
+        # per-layer all-reduce counts and deadlock the group).  At
+        # world=1 there is no broadcast, so the normal inbox admit must
+        # stay on (found by the TP=1 GPU rehearsal: requests were never
+        # admitted and generate_sync timed out empty).
+        self._tp_direct = tp.world > 1
         # group watchdog: model keys pending a synchronized reset
         self._reset_pending: set = set()
         self._model_order = sorted(self.models)

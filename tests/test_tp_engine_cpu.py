@@ -123,3 +123,32 @@ def test_tp_group_watchdog_resets_all_ranks_together():
     assert results["r0"][0] is None, results["r0"][0]
     # BOTH ranks rebuilt their BlockManager (not just the faulty one)
     assert results["r0"][1] and results["r1"][1], results
+
+
+def test_tp_engine_world1_serves_requests():
+    """TP=1 TPEngine (config-5 mechanism at trivial world) must serve
+    through the normal inbox — found by the GPU rehearsal: _tp_direct
+    suppressed admission entirely at world=1."""
+    import torch.distributed as dist
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ["MASTER_PORT"] = "29553"
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        from quoracle_amd.engine.api import GenerateRequest
+        from quoracle_amd.engine.tp_engine import TPEngine
+        from quoracle_amd.parallel.tp import TPContext
+        eng = TPEngine(["tiny#tp1"], TPContext(0, 1),
+                       device=torch.device("cpu"), kv_blocks_override=256,
+                       embed_model_key=None, prefill_chunk=64).start()
+        try:
+            r = eng.generate_sync(GenerateRequest(
+                model_key="tiny#tp1",
+                messages=[{"role": "user", "content": "world one"}],
+                temperature=0.7, max_tokens=120, seed=9,
+                action_grammar=True, session_id="w1"), timeout=120)
+            assert r.ok, r.error
+            assert r.output_tokens > 0
+        finally:
+            eng.stop()
+    finally:
+        dist.destroy_process_group()
