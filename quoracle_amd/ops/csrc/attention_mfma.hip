@@ -1515,3 +1515,240 @@ paged_attn_prefill_t12_split_kernel(
           dead ? 0.f : o_acc[db][r];
   }
 }
+
+// ---------------------------------------------------------------------------
+// EXPERIMENTAL T12W prefill: the T12 structure on the 32x32x16 MFMA
+// (double the flops per instruction: guide µbench 2382 vs 2075 TF).
+// 4 waves per WG, wave owns a 32-row qblock (same 128-row tiles as T12).
+// Swapped QK^T gives lane-local S rows (qrow = lane&31, keys by the
+// 32x32 C map (reg&3)+8*(reg>>2)+4*(lane>>5)); softmax fully in
+// registers (one shfl_xor(32)); the PV A-fragments come from FOUR
+// permlane32_swaps per 32-key block:
+//   swap(j0,j2) -> [ks=0 D0, ks=0 D2]; swap(j1,j3) -> [D1, D3]
+//   swap(j4,j6) -> [ks=1 D0, D2];      swap(j5,j7) -> [D1, D3]
+// (dword j = cvt_pk of P at regs 2j, 2j+1 — adjacent keys by the C map).
+// Q per wave lives in q_s (XOR-swizzled [128][128]); K/V staged as in
+// the 4-wave kernels; 2 barriers per 64-key chunk.
+// Grid: (ntiles128, Hq); block 256.
+// ---------------------------------------------------------------------------
+typedef float f32x16_t __attribute__((ext_vector_type(16)));
+
+extern "C" __global__ void __launch_bounds__(256)
+paged_attn_prefill_t12w_kernel(
+    bf16 *__restrict__ out, const bf16 *__restrict__ q,
+    const bf16 *__restrict__ kc, const bf16 *__restrict__ vc,
+    const int *__restrict__ bt, const int *__restrict__ tile_q0,
+    const int *__restrict__ tile_qn, const int *__restrict__ tile_seq,
+    const int *__restrict__ tile_pos0, float scale, int Hq, int Hkv, int BS,
+    int MAXB, int GQ) {
+  const int tile = blockIdx.x;
+  const int h = blockIdx.y;
+  const int hk = h / GQ;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int q0 = tile_q0[tile];
+  const int qn = tile_qn[tile];
+  const int seq = tile_seq[tile];
+  const int pos0 = tile_pos0[tile];
+  const int kv_limit = pos0 + qn;
+
+  __shared__ bf16 q_s[T12_QT * MF_D];        // XOR-swizzled 128-row Q
+  __shared__ bf16 k_s[MF_KCHUNK * MF_D];     // XOR-swizzled
+  __shared__ bf16 v_s[MF_KCHUNK * VR];       // row-major
+  __shared__ float alpha_w[4][32];
+  __shared__ float l_w[4][32];
+
+  const int qr = lane & 31;                  // lane's qrow within qblock
+  const int hi = lane >> 5;
+  const int row_local = wave * 32 + qr;      // row within the 128-row tile
+
+  // stage the Q tile (swizzled, all 4 waves cooperate)
+  for (int i = tid; i < T12_QT * MF_D / 8; i += 256) {
+    const int r = (i * 8) / MF_D, c = (i * 8) % MF_D;
+    uint4 val = make_uint4(0, 0, 0, 0);
+    if (r < qn)
+      val = reinterpret_cast<const uint4 *>(
+          q + ((long)(q0 + r) * Hq + h) * MF_D + c)[0];
+    reinterpret_cast<uint4 *>(
+        q_s + r * MF_D + (c ^ ((r & 15) << 3)))[0] = val;
+  }
+
+  float m_run = -INFINITY, l_run = 0.f;
+  f32x16_t o_acc[4];                         // 4 d-tiles of 32 cols
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt) o_acc[dt] = (f32x16_t){};
+
+  const long panel_stride = (long)Hkv * BS * MF_D;
+  const int abs_qrow = min(row_local, qn - 1) + pos0;
+
+  uint4 kreg[4], vreg[4];
+  auto issue_loads = [&](int start_, int limit_) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      uint4 kv = make_uint4(0, 0, 0, 0), vv = make_uint4(0, 0, 0, 0);
+      const int token = start_ + key;
+      if (token < limit_) {
+        const long blk = bt[(long)seq * MAXB + token / BS];
+        const long off =
+            blk * panel_stride + ((long)hk * BS + token % BS) * MF_D + d;
+        kv = reinterpret_cast<const uint4 *>(kc + off)[0];
+        vv = reinterpret_cast<const uint4 *>(vc + off)[0];
+      }
+      kreg[it] = kv;
+      vreg[it] = vv;
+    }
+  };
+  auto write_staged = [&]() {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int i = tid + it * 256;
+      const int key = (i * 8) / MF_D, d = (i * 8) % MF_D;
+      reinterpret_cast<uint4 *>(
+          k_s + key * MF_D + (d ^ ((key & 15) << 3)))[0] = kreg[it];
+      const uint *vw = reinterpret_cast<const uint *>(&vreg[it]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        reinterpret_cast<uint *>(v_s + key * VR + d)[j] = vw[j];
+    }
+  };
+
+  auto packbf = [](float lo, float hif) {
+    union { bf16 hh; unsigned short u; } a, b;
+    a.hh = f2bf(lo);
+    b.hh = f2bf(hif);
+    return (uint)a.u | ((uint)b.u << 16);
+  };
+
+  issue_loads(0, kv_limit);
+  for (int start = 0; start < kv_limit; start += MF_KCHUNK) {
+    const int clen = min(MF_KCHUNK, kv_limit - start);
+    write_staged();
+    __syncthreads();
+    if (start + MF_KCHUNK < kv_limit)
+      issue_loads(start + MF_KCHUNK, kv_limit);
+
+    // ---- S^T = K·Q^T: 2 key-blocks of 32, k-loop over D in 16s -------
+    f32x16_t s_acc[2];
+    s_acc[0] = (f32x16_t){};
+    s_acc[1] = (f32x16_t){};
+#pragma unroll
+    for (int kk = 0; kk < MF_D / 16; ++kk) {
+      const int doff = (kk * 16 + hi * 8);
+      bf16x8_t qf = *reinterpret_cast<const bf16x8_t *>(
+          q_s + (wave * 32 + qr) * MF_D
+          + (doff ^ (((wave * 32 + qr) & 15) << 3)));
+#pragma unroll
+      for (int kb = 0; kb < 2; ++kb) {
+        bf16x8_t kf = *reinterpret_cast<const bf16x8_t *>(
+            k_s + (kb * 32 + qr) * MF_D + (doff ^ ((qr & 15) << 3)));
+        s_acc[kb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf,
+                                                            s_acc[kb], 0, 0, 0);
+      }
+    }
+
+    // ---- in-register online softmax over the 64-key chunk ------------
+    float mymax = -INFINITY;
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int key = kb * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int token = start + key;
+        const bool ok = (token <= abs_qrow) && (key < clen);
+        const float sv = ok ? s_acc[kb][r] * scale : -INFINITY;
+        s_acc[kb][r] = sv;
+        mymax = fmaxf(mymax, sv);
+      }
+    }
+    mymax = fmaxf(mymax, __shfl_xor(mymax, 32));
+    const float mn = fmaxf(m_run, mymax);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - mn);
+    float psum = 0.f;
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float p = (s_acc[kb][r] == -INFINITY || mn == -INFINITY)
+                            ? 0.f : __expf(s_acc[kb][r] - mn);
+        psum += p;
+        s_acc[kb][r] = p;
+      }
+    }
+    psum += __shfl_xor(psum, 32);
+    l_run = l_run * alpha + psum;
+    m_run = mn;
+    if (lane < 32) alpha_w[wave][lane] = alpha;   // qrow == lane here
+
+    // ---- O rescale (alphas for the C row map via the wave strip) -----
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        o_acc[dt][r] *= alpha_w[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
+    }
+
+    // ---- P -> A fragments (4 swaps per 32-key block) + PV ------------
+#pragma unroll
+    for (int kb = 0; kb < 2; ++kb) {
+      uint j[8];
+#pragma unroll
+      for (int m = 0; m < 8; ++m)
+        j[m] = packbf(s_acc[kb][2 * m], s_acc[kb][2 * m + 1]);
+      // swap(j0,j2): ks=0 D0/D2; swap(j1,j3): ks=0 D1/D3;
+      // swap(j4,j6): ks=1 D0/D2; swap(j5,j7): ks=1 D1/D3
+      uint a0d0, a0d1, a0d2, a0d3, a1d0, a1d1, a1d2, a1d3;
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(j[0], j[2], false, false);
+        a0d0 = r[0]; a0d2 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(j[1], j[3], false, false);
+        a0d1 = r[0]; a0d3 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(j[4], j[6], false, false);
+        a1d0 = r[0]; a1d2 = r[1];
+      }
+      {
+        auto r = __builtin_amdgcn_permlane32_swap(j[5], j[7], false, false);
+        a1d1 = r[0]; a1d3 = r[1];
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const uint4 av = half
+            ? make_uint4(a1d0, a1d1, a1d2, a1d3)
+            : make_uint4(a0d0, a0d1, a0d2, a0d3);
+        const bf16x8_t a_frag = __builtin_bit_cast(bf16x8_t, av);
+        const int ksbase = kb * 32 + half * 16;      // key-slice origin
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          bf16x8_t b;
+#pragma unroll
+          for (int t = 0; t < 8; ++t)
+            b[t] = v_s[(ksbase + hi * 8 + t) * VR + dt * 32 + qr];
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_frag, b, o_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();   // next chunk restages k_s / v_s
+  }
+
+  if (lane < 32) l_w[wave][lane] = l_run;
+  // epilogue: C rows = (r&3)+8*(r>>2)+4*hi of this wave's qblock
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int rowq = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const int row = wave * 32 + rowq;
+    if (row >= qn) continue;
+    const float lv = l_w[wave][rowq];
+    const float denom = lv > 0.f ? lv : 1.f;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      out[((long)(q0 + row) * Hq + h) * MF_D + dt * 32 + qr] =
+          f2bf(o_acc[dt][r] / denom);
+  }
+}

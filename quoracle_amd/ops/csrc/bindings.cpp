@@ -524,6 +524,37 @@ void paged_attn_prefill_mfma32(torch::Tensor out, torch::Tensor q,
   HIP_CHECK_KERNEL();
 }
 
+void paged_attn_prefill_t12w(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor kcache, torch::Tensor vcache,
+                             torch::Tensor block_tables,
+                             torch::Tensor tile_q0, torch::Tensor tile_qn,
+                             torch::Tensor tile_seq,
+                             torch::Tensor tile_pos0, double scale) {
+  CHECK_GPU(out);
+  CHECK_GPU(q);
+  CHECK_GPU(kcache);
+  CHECK_GPU(vcache);
+  CHECK_GPU(block_tables);
+  const int ntiles = tile_q0.size(0);
+  const int Hq = q.size(1);
+  const int D = q.size(2);
+  const int Hkv = kcache.size(1);
+  const int BS = kcache.size(2);
+  const int MAXB = block_tables.size(1);
+  const int GQ = Hq / Hkv;
+  TORCH_CHECK(D == 128, "T12W prefill kernel requires head dim 128");
+  TORCH_CHECK(Hq % Hkv == 0, "bad GQA ratio");
+  if (ntiles == 0) return;
+  hipLaunchKernelGGL(paged_attn_prefill_t12w_kernel, dim3(ntiles, Hq),
+                     dim3(256), 0, current_stream(), bf16_ptr(out),
+                     bf16_cptr(q), bf16_cptr(kcache), bf16_cptr(vcache),
+                     block_tables.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                     tile_qn.data_ptr<int>(), tile_seq.data_ptr<int>(),
+                     tile_pos0.data_ptr<int>(), (float)scale, Hq, Hkv, BS,
+                     MAXB, GQ);
+  HIP_CHECK_KERNEL();
+}
+
 void paged_attn_prefill_t12(torch::Tensor out, torch::Tensor q,
                             torch::Tensor kcache, torch::Tensor vcache,
                             torch::Tensor block_tables,
@@ -613,6 +644,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Context-split prefill attention + combine (small chunks)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
         "MFMA-tiled prefill attention (D=128, matrix cores)");
+  m.def("paged_attn_prefill_t12w", &paged_attn_prefill_t12w,
+        "EXPERIMENTAL: T12 structure on the 32x32x16 MFMA (validate "
+        "before use)");
   m.def("paged_attn_prefill_t12_split", &paged_attn_prefill_t12_split,
         "EXPERIMENTAL: context-split T12 prefill, wave-local chunks "
         "(partials over NS*8 splits) + combine");

@@ -151,3 +151,9 @@ if hasattr(ops.ext(), "paged_attn_prefill_t12_split"):
     rel_s = (osc.float() - o_spl.float()).norm() / o_spl.float().norm()
     print(f"t12_split small-chunk (EXPERIMENTAL): {ms_t12s*1e3:.1f} us "
           f"({ms_spl/ms_t12s:.2f}x)  rel-vs-mfma_split {rel_s:.4f}")
+if hasattr(ops.ext(), "paged_attn_prefill_t12w"):
+    ms_t12w = timeit(lambda: ops.ext().paged_attn_prefill_t12w(
+        out, q, kcache, vcache, tables, t128, qn128, ts128, tp128, scale))
+    rel12w = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
+    print(f"T12W prefill (EXPERIMENTAL 32x32): {ms_t12w:.3f} ms "
+          f"({flops/ms_t12w/1e9:.1f} TFLOP/s)  rel-vs-16 {rel12w:.4f}")

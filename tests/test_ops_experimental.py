@@ -66,3 +66,29 @@ def test_paged_attn_prefill_t12_split_matches_reference(dev):
     ref = reference.attention(q, k, v, scale, causal_offset=cached)
     assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
         f"max err {(out.float() - ref).abs().max().item()}"
+
+
+def test_paged_attn_prefill_t12w_matches_reference(dev):
+    """T12W (32x32x16 MFMA, 4-swap permlane A-fragments) vs the fp32
+    reference, incl. an odd tail."""
+    ops = _ops()
+    torch.manual_seed(31)
+    Hq, Hkv, D, BS = 32, 8, 128, 16
+    scale = D ** -0.5
+    cached, new = 777, 300
+    total = cached + new
+    k = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(total, Hkv, D, device=dev, dtype=torch.bfloat16)
+    kcache, vcache, tables, ctx = _build_paged_cache(dev, [(k, v)], Hkv, D, BS)
+    q = torch.randn(new, Hq, D, device=dev, dtype=torch.bfloat16)
+    out = torch.empty_like(q)
+    nt = (new + 127) // 128
+    t0 = torch.arange(nt, dtype=torch.int32, device=dev) * 128
+    qn = torch.clamp(torch.full_like(t0, new) - t0, max=128)
+    tseq = torch.zeros_like(t0)
+    tpos = t0 + cached
+    ops.ext().paged_attn_prefill_t12w(out, q, kcache, vcache, tables,
+                                      t0, qn, tseq, tpos, scale)
+    ref = reference.attention(q, k, v, scale, causal_offset=cached)
+    assert torch.allclose(out.float(), ref, atol=4e-2, rtol=4e-2), \
+        f"max err {(out.float() - ref).abs().max().item()}"
