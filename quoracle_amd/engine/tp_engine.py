@@ -67,6 +67,11 @@ class TPEngine(LocalEngine):
 
     def request_stop(self) -> None:
         """Leader: broadcast shutdown to replicas on the next step."""
+        if self.tp.world <= 1:
+            # no broadcast at world 1 — the sentinel would land in the
+            # NORMAL admit path, which expects sequences
+            self.stop_seen = True
+            return
         self._inbox.put(_StopSentinel())
 
     def reset_model(self, key: str) -> None:
