@@ -157,3 +157,19 @@ if hasattr(ops.ext(), "paged_attn_prefill_t12w"):
     rel12w = (out.float() - out_mfma.float()).norm() / out_mfma.float().norm()
     print(f"T12W prefill (EXPERIMENTAL 32x32): {ms_t12w:.3f} ms "
           f"({flops/ms_t12w/1e9:.1f} TFLOP/s)  rel-vs-16 {rel12w:.4f}")
+# --- decode NS sensitivity (pick the split-count formula) ------------------
+if hasattr(ops.ext(), "paged_attn_decode_mfma"):
+    for NSx in (8, 16, 27, 40, 56):
+        pmx = torch.empty((B, Hq, NSx), dtype=torch.float32, device=dev)
+        plx = torch.empty_like(pmx)
+        pax = torch.empty((B, Hq, NSx, D), dtype=torch.float32, device=dev)
+        msx = timeit(lambda: ops.ext().paged_attn_decode_mfma(
+            outd, qd, kcache, vcache, tables2, ctxs, scale,
+            pmx, plx, pax), n=50)
+        m1x = timeit(lambda: ops.ext().paged_attn_decode_mfma(
+            outd[:1].contiguous(), qd[:1].contiguous(), kcache, vcache,
+            tables2[:1].contiguous(), ctxs[:1].contiguous(), scale,
+            pmx[:1].contiguous(), plx[:1].contiguous(),
+            pax[:1].contiguous()), n=50)
+        print(f"decode_mfma NS={NSx}: B=8 {msx*1e3:.1f} us "
+              f"({kv_bytes/msx/1e9:.2f} TB/s)  B=1 {m1x*1e3:.1f} us")
