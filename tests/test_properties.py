@@ -572,3 +572,60 @@ def test_ui_paths_never_500(path_bit):
                     f"/api/tasks/{pb}/export"):
             r = client.get(url)
             assert r.status_code < 500, (url, r.status_code, r.text[:200])
+
+
+_ACTION_JSONS = st.sampled_from([
+    '{"reasoning": "r", "action": "wait", "params": {"wait": true}}',
+    '{"reasoning": "r", "action": "orient", "params": {"current_situation":'
+    ' "s", "goal_clarity": "g", "available_resources": "a",'
+    ' "key_challenges": "k", "delegation_consideration": "d"}}',
+    '{"reasoning": "r", "action": "send_message", "params":'
+    ' {"to": "parent", "content": "hello"}}',
+    '{"action": "todo", "params": {"items": [{"content": "x",'
+    ' "state": "todo"}]}}',
+    '{"action": "record_cost", "params": {"amount": "3"}}',
+])
+
+
+@settings(max_examples=40, deadline=None)
+@given(scripts=st.lists(
+    st.lists(st.one_of(_ACTION_JSONS, st.text(max_size=60), st.none()),
+             min_size=1, max_size=6),
+    min_size=1, max_size=3),
+    rounds=st.integers(min_value=0, max_value=4))
+def test_consensus_pipeline_total_over_arbitrary_responses(scripts, rounds):
+    """The full consensus loop (clustering -> unanimity/majority ->
+    refinement -> forced decision) over ARBITRARY per-model response
+    streams (valid actions, junk text, model failures): terminates within
+    the round budget with exactly one decision or a structured
+    ConsensusError — never hangs, never crashes."""
+    import asyncio as _asyncio
+    from quoracle_amd.consensus import pipeline as P
+
+    pool = [f"m{i}" for i in range(len(scripts))]
+    calls = {m: 0 for m in pool}
+
+    async def query_fn(model_key, round_num, refinement_prompt):
+        i = pool.index(model_key)
+        seq = scripts[i]
+        out = seq[min(calls[model_key], len(seq) - 1)]
+        calls[model_key] += 1
+        if out is None:
+            raise RuntimeError("model failure")
+        return out
+
+    async def drive():
+        try:
+            outcome = await _asyncio.wait_for(P.run_consensus(
+                pool, query_fn, max_refinement_rounds=rounds), timeout=30)
+            assert outcome.decision.action
+            assert 1 <= outcome.rounds_used <= rounds + 1
+            return True
+        except P.ConsensusError as exc:
+            assert exc.reason in ("all_models_failed",
+                                  "all_responses_invalid")
+            return False
+
+    _asyncio.new_event_loop().run_until_complete(drive())
+    # no model was queried more than once per round (+1 forced round)
+    assert all(c <= rounds + 1 for c in calls.values()), calls
