@@ -346,3 +346,45 @@ def test_registry_duplicate_id_raises():
         reg.register("dup", object(), "t1")
     reg.unregister("dup")
     reg.register("dup", object(), "t1")   # reusable after unregister
+
+
+def test_temperature_schedule_golden_table():
+    """Full golden table for the linear descent (reference:
+    consensus/temperature.ex): intermediate rounds, past-max clamping,
+    degenerate max_rounds, provider-prefixed names."""
+    cases = [
+        # (model, round, max_rounds, expected) — one-decimal rounding is
+        # part of the contract (temperature.ex rounds the same way)
+        ("gpt-4o", 2, 4, 1.5),
+        ("gpt-4o", 3, 4, 0.9),
+        ("llama3-8b#1", 2, 4, 0.7),
+        ("llama3-8b#1", 3, 4, 0.5),
+        ("llama3-8b#1", 9, 4, 0.2),          # past max clamps at floor
+        ("gpt-4o", 9, 4, 0.4),
+        ("openai:gpt-4o", 1, 4, 2.0),        # provider-prefixed names
+        ("gemini-pro", 1, 4, 2.0),
+        ("o1-preview", 1, 4, 2.0),
+        ("mixtral-8x7b#0", 1, 9, 1.0),
+        ("mixtral-8x7b#0", 9, 9, 0.2),
+        ("mixtral-8x7b#0", 5, 9, 0.6),
+    ]
+    for model, rnd, mx, want in cases:
+        got = temp.round_temperature(model, rnd, mx)
+        assert got == pytest.approx(want, abs=1e-9), (model, rnd, mx, got)
+    # max_rounds <= 1: single round at max temperature, no division blowup
+    assert temp.round_temperature("llama3-8b#1", 1, 1) == 1.0
+    assert temp.round_temperature("llama3-8b#1", 1, 0) == 1.0
+
+
+def test_confidence_golden_boundaries():
+    from quoracle_amd.consensus.result import calculate_confidence
+    # bonus tier boundaries are EXCLUSIVE (> not >=)
+    assert calculate_confidence(1, 2, 2) == pytest.approx(0.5)       # 0.5 -> no bonus
+    assert calculate_confidence(3, 5, 2) == pytest.approx(0.6 + 0.05)  # >0.5 tier
+    assert calculate_confidence(6, 10, 2) == pytest.approx(0.6 + 0.05)
+    assert calculate_confidence(4, 5, 2) == pytest.approx(0.8 + 0.10)  # 0.8 -> mid tier
+    assert calculate_confidence(9, 10, 2) == 1.0   # 0.9+0.15 clamped
+    # late-round penalty and the [0.1, 1.0] clamp
+    assert calculate_confidence(1, 4, 7, 4) == pytest.approx(0.1)
+    assert calculate_confidence(1, 10, 9, 4) == 0.1
+    assert calculate_confidence(10, 10, 1) == 1.0
