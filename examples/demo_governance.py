@@ -9,6 +9,11 @@ write.  Run: python examples/demo_governance.py
 """
 
 import asyncio
+import os
+
+# the vault refuses to exist without key material (docs/SECURITY.md);
+# a deployment sets this in its environment
+os.environ.setdefault("QUORACLE_VAULT_KEY", "demo-governance-key")
 import json
 import os
 import sys
