@@ -173,3 +173,37 @@ if hasattr(ops.ext(), "paged_attn_decode_mfma"):
             pax[:1].contiguous()), n=50)
         print(f"decode_mfma NS={NSx}: B=8 {msx*1e3:.1f} us "
               f"({kv_bytes/msx/1e9:.2f} TB/s)  B=1 {m1x*1e3:.1f} us")
+# --- context-scaling sweep (QUORACLE_ATTN_SWEEP=1) -------------------------
+if os.environ.get("QUORACLE_ATTN_SWEEP"):
+    for ctxs_len in (2000, 7000, 16000, 32000):
+        nbS = (ctxs_len + BS - 1) // BS + 1
+        kcS = torch.randn(nbS, Hkv, BS, D, device=dev, dtype=torch.bfloat16)
+        vcS = torch.randn_like(kcS)
+        tabS = torch.stack([torch.arange(1, nbS, dtype=torch.int32,
+                                         device=dev) for _ in range(8)])
+        qS = torch.randn(8, Hq, D, device=dev, dtype=torch.bfloat16)
+        oS = torch.empty_like(qS)
+        cS = torch.full((8,), ctxs_len, dtype=torch.int32, device=dev)
+        NSS = min(32, max(2, ctxs_len // 256))
+        pmS = torch.empty((8, Hq, NSS), dtype=torch.float32, device=dev)
+        plS = torch.empty_like(pmS)
+        paS = torch.empty((8, Hq, NSS, D), dtype=torch.float32, device=dev)
+        msS = timeit(lambda: ops.ext().paged_attn_decode_mfma(
+            oS, qS, kcS, vcS, tabS, cS, scale, pmS, plS, paS), n=30)
+        kvb = 2 * 8 * ctxs_len * Hkv * D * 2
+        # T12 prefill: 1024 new tokens over this cached context
+        newS = 1024
+        qP = torch.randn(newS, Hq, D, device=dev, dtype=torch.bfloat16)
+        oP = torch.empty_like(qP)
+        ntS = (newS + 127) // 128
+        t0S = torch.arange(ntS, dtype=torch.int32, device=dev) * 128
+        qnS = torch.clamp(torch.full_like(t0S, newS) - t0S, max=128)
+        tsS = torch.zeros_like(t0S)
+        tpS = t0S + ctxs_len - newS
+        msP = timeit(lambda: ops.ext().paged_attn_prefill_t12(
+            oP, qP, kcS, vcS, tabS[:1].contiguous(), t0S, qnS, tsS, tpS,
+            scale), n=10)
+        flP = 2 * 2 * newS * (ctxs_len - newS / 2) * D * Hq
+        print(f"ctx={ctxs_len:6d}: decode_mfma B=8 {msS*1e3:7.1f} us "
+              f"({kvb/msS/1e9:.2f} TB/s)   t12 prefill 1024-tok chunk "
+              f"{msP:7.3f} ms ({flP/msP/1e9:.0f} TF)")
