@@ -97,12 +97,13 @@ async def run(args):
     Supervisor(runtime)
     manager = TaskManager(runtime)
 
-    # The coordinator-side logic (spawn solvers, collect, score, write)
-    # is exercised deterministically: one solver child per subject is
-    # spawned via the REAL spawn action path, each question is put to
-    # the child as a user message, the child's consensus cycle answers
-    # through the REAL pipeline, and the score file passes the grove's
-    # write schema.
+    # Harness surfaces exercised for real: task creation with the grove,
+    # one solver child per subject via the REAL spawn action path (and
+    # dismissal after), per-question answering through the engine
+    # protocol, and the score file through the confined schema-validated
+    # write.  (The per-question loop drives the engine directly rather
+    # than a full consensus cycle per question — 600 scripted consensus
+    # cycles add nothing over the covered cycle tests.)
     result = await manager.create_task(
         "Run the QA benchmark", "default", grove=grove)
     root = runtime.registry.lookup(result["root_agent_id"]).actor
