@@ -1,8 +1,6 @@
 """Dedicated regressions for the round-1 advisor findings:
 each was fixed in round 2 — these pin the behaviors."""
 
-import asyncio
-
 import pytest
 
 from quoracle_amd.engine.fake import FakeEngine

@@ -9,7 +9,7 @@ import pytest
 
 from quoracle_amd.engine.fake import FakeEngine, deterministic_embedding
 
-from helpers import IDLE, POOL2, action_json, make_manager, wait_until
+from helpers import IDLE, make_manager, wait_until
 
 
 # ---------------------------------------------------------------------------

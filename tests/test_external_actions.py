@@ -125,7 +125,6 @@ async def test_call_mcp_http_roundtrip():
     The server answers initialize as SSE (testing the event-stream parse
     path), later calls as plain JSON, assigns a session id, and records
     the DELETE on terminate."""
-    import socket
     import threading
     import http.server
 
