@@ -325,7 +325,11 @@ class LlamaModel:
         S = T * K
         expert_flat = experts.reshape(-1)
         e_sorted = expert_flat[order]
-        counts = torch.bincount(expert_flat, minlength=E)
+        # scatter_add, not bincount: bincount syncs to size its output,
+        # which breaks hipGraph capture of MoE decode steps
+        counts = torch.zeros(E, dtype=expert_flat.dtype,
+                             device=expert_flat.device)
+        counts.scatter_add_(0, expert_flat, torch.ones_like(expert_flat))
         raw_off = torch.cumsum(counts, 0) - counts             # exclusive
         pos = torch.arange(S, device=h.device) - raw_off[e_sorted]
         idx = e_sorted * T + pos                               # bin slots

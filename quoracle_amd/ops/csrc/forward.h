@@ -191,7 +191,11 @@ inline torch::Tensor moe_ffn_bmm(const Model &m, const Layer &L,
   auto order = expert_flat.argsort(/*stable=*/true);
   auto e_sorted = expert_flat.index_select(0, order);
   auto token_of = order.div(K, "floor");
-  auto counts = at::bincount(expert_flat, /*weights=*/{}, /*minlength=*/E);
+  // counts via scatter_add, NOT at::bincount: bincount computes its
+  // output size with a device->host sync, which is illegal inside
+  // hipGraph capture (this was the dual-preset capture failure)
+  auto counts = torch::zeros({E}, expert_flat.options());
+  counts.scatter_add_(0, expert_flat, torch::ones_like(expert_flat));
   auto raw_off = at::cumsum(counts, 0) - counts;   // exclusive prefix
   auto pos = at::arange(S, order.options()) -
              raw_off.index_select(0, e_sorted);
