@@ -1093,7 +1093,8 @@ paged_attn_decode_mfma_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// EXPERIMENTAL T12 prefill: 128-row Q tiles, 8 waves, each wave OWNS one
+// T12 prefill (the DEFAULT big-prefill kernel, 352 TF measured on the
+// bench shape): 128-row Q tiles, 8 waves, each wave OWNS one
 // 16-row qblock end to end.  Swapped QK^T (mfma(K, Q)) makes every S row
 // lane-local, so the online softmax runs fully in registers (2 shfl_xor
 // across the 4 same-qrow lanes) — no s_s/p_s LDS round trip and only TWO

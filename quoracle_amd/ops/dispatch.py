@@ -72,7 +72,7 @@ def paged_attn_decode(out, q, kcache, vcache, block_tables, ctx_lens,
             part_acc = torch.empty((B, Hq, ns, D), dtype=torch.float32,
                                    device=q.device)
             if D == 128 and not os.environ.get("QUORACLE_DECODE_VALU"):
-                # matrix-core flash-decode (validated r2: 0.78 -> ~3 TB/s)
+                # matrix-core flash-decode (validated r2: 0.78 -> 4.3 TB/s at 7k ctx, 6.1 at 32k)
                 ext().paged_attn_decode_mfma(out, q, kcache, vcache,
                                              block_tables, ctx_lens, scale,
                                              part_m, part_l, part_acc)
